@@ -1,0 +1,25 @@
+"""mi355ml — MI355X-native distributed classical ML.
+
+A from-scratch framework with the capabilities of the dask-ml lineage
+(sklearn-compatible .fit/.transform/.predict estimators, joblib persistence),
+re-designed MI355X-first: row-sharded device arrays (one process per GPU),
+RCCL collectives over xGMI instead of task-graph tree-reductions, and
+hand-written CDNA4 HIP kernels for the hot per-shard math.
+
+Reference capability map: SURVEY.md (mrocklin/dask-ml lineage).
+"""
+
+__version__ = "0.1.0"
+
+from . import config
+from .core.array import DistArray, from_numpy, shard_rows
+from .parallel import comm
+
+__all__ = [
+    "config",
+    "comm",
+    "DistArray",
+    "from_numpy",
+    "shard_rows",
+    "__version__",
+]

@@ -1,0 +1,96 @@
+"""Estimator base classes: sklearn API contract + joblib persistence.
+
+Estimators follow the sklearn contract the reference mirrors
+(sklearn/base.py:141,178,202 get_params/set_params; :668 fit_transform;
+:470 score) by subclassing sklearn.base.BaseEstimator — same choice as
+dask-ml. Fitted attributes live as small replicated torch tensors on the
+rank's device; `__getstate__` moves them to host numpy so `joblib.dump`
+produces portable files (SURVEY §2.1 model-persistence contract).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Union
+
+import numpy as np
+import torch
+from sklearn.base import BaseEstimator as _SkBase
+from sklearn.base import ClassifierMixin, RegressorMixin, TransformerMixin  # noqa: F401
+
+from .core.array import DistArray, from_numpy, shard_rows
+from .parallel import comm as comm_mod
+
+ArrayLike = Union[DistArray, np.ndarray, torch.Tensor]
+
+
+def check_array(X: ArrayLike, dtype: torch.dtype = torch.float32,
+                comm=None) -> DistArray:
+    """Coerce input to a row-sharded DistArray (dask-tolerant check_array
+    analog, dask_ml/utils.py)."""
+    if isinstance(X, DistArray):
+        if dtype is not None and X.dtype != dtype:
+            return X.astype(dtype)
+        return X
+    if isinstance(X, np.ndarray):
+        t = torch.from_numpy(np.ascontiguousarray(X))
+    elif isinstance(X, torch.Tensor):
+        t = X
+    else:
+        t = torch.as_tensor(np.asarray(X))
+    if dtype is not None and t.dtype != dtype:
+        t = t.to(dtype)
+    return shard_rows(t, comm)
+
+
+def check_X_y(X: ArrayLike, y: ArrayLike, dtype: torch.dtype = torch.float32,
+              y_dtype: torch.dtype = torch.float32, comm=None):
+    Xd = check_array(X, dtype, comm)
+    yd = check_array(y, y_dtype, Xd.comm)
+    if Xd.global_rows != yd.global_rows:
+        raise ValueError(
+            f"X and y have inconsistent lengths: {Xd.global_rows} vs "
+            f"{yd.global_rows}")
+    return Xd, yd
+
+
+def check_is_fitted(est, attr: str) -> None:
+    if not hasattr(est, attr):
+        raise RuntimeError(
+            f"{type(est).__name__} is not fitted; call fit() first.")
+
+
+def _to_host(v):
+    if isinstance(v, torch.Tensor):
+        return ("__mi355ml_tensor__", v.detach().cpu().numpy())
+    return v
+
+
+def _from_host(v, device):
+    if isinstance(v, tuple) and len(v) == 2 and v[0] == "__mi355ml_tensor__":
+        return torch.from_numpy(np.ascontiguousarray(v[1])).to(device)
+    return v
+
+
+class BaseEstimator(_SkBase):
+    """sklearn-compatible base with device-aware pickling."""
+
+    def __getstate__(self):
+        state = self.__dict__.copy()
+        return {k: _to_host(v) for k, v in state.items()}
+
+    def __setstate__(self, state):
+        device = comm_mod.default().device
+        self.__dict__.update(
+            {k: _from_host(v, device) for k, v in state.items()})
+
+    # convenience for subclasses
+    def _device(self) -> torch.device:
+        return comm_mod.default().device
+
+
+def as_result(t: torch.Tensor, like: ArrayLike):
+    """Return small fitted/derived results in a caller-friendly form:
+    numpy in, numpy out; tensors pass through."""
+    if isinstance(like, np.ndarray):
+        return t.detach().cpu().numpy()
+    return t

@@ -1,0 +1,226 @@
+"""DistArray — eager row-sharded device array.
+
+The MI355X replacement for dask.array.Array (dask/array/core.py:1137): instead
+of a lazy task graph over chunks, a DistArray is ONE resident torch tensor per
+rank (row shard), sized for 288 GB HBM3E, operated on SPMD-style. Every
+cross-chunk pattern in the reference's graphs maps to one collective
+(SURVEY §2.7 C-table):
+
+  reference graph pattern                here
+  -------------------------------------  -----------------------------------
+  _tree_reduce of column stats (C1)       ops.col_* + allgather/allreduce
+  tensordot + sum tree (C2)               local GEMM + allreduce
+  concatenate/gather of blocks (C5)       all_gather(_varlen)
+  rechunk/shuffle (C9)                    all_to_all (not needed for row ops)
+
+Rows are split contiguously and near-evenly: rank r owns [lo, hi) from
+Communicator.shard_bounds. 1-D arrays (labels/targets) shard the same way.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Iterator, Optional, Tuple
+
+import numpy as np
+import torch
+
+from .. import config
+from ..parallel import comm as comm_mod
+from ..parallel.comm import Communicator
+from .. import ops
+
+
+class DistArray:
+    """Row-sharded dense array (1-D or 2-D), eager, one shard per rank."""
+
+    def __init__(self, local: torch.Tensor, global_rows: int, row_offset: int,
+                 comm: Optional[Communicator] = None):
+        assert local.dim() in (1, 2), "DistArray is 1-D or 2-D"
+        self.local = local
+        self.global_rows = int(global_rows)
+        self.row_offset = int(row_offset)
+        self.comm = comm if comm is not None else comm_mod.default()
+
+    # ------------------------------------------------------------- basics
+    @property
+    def shape(self) -> Tuple[int, ...]:
+        if self.local.dim() == 1:
+            return (self.global_rows,)
+        return (self.global_rows, self.local.shape[1])
+
+    @property
+    def ndim(self) -> int:
+        return self.local.dim()
+
+    @property
+    def ncols(self) -> int:
+        return 1 if self.local.dim() == 1 else self.local.shape[1]
+
+    @property
+    def dtype(self) -> torch.dtype:
+        return self.local.dtype
+
+    @property
+    def device(self) -> torch.device:
+        return self.local.device
+
+    def __len__(self) -> int:
+        return self.global_rows
+
+    def __repr__(self) -> str:  # pragma: no cover
+        return (f"DistArray(shape={self.shape}, dtype={self.dtype}, "
+                f"device={self.device}, rank={self.comm.rank}/"
+                f"{self.comm.world_size}, local_rows={self.local.shape[0]})")
+
+    def _like(self, local: torch.Tensor) -> "DistArray":
+        assert local.shape[0] == self.local.shape[0]
+        return DistArray(local, self.global_rows, self.row_offset, self.comm)
+
+    # ---------------------------------------------------------- transforms
+    def map_local(self, fn: Callable[[torch.Tensor], torch.Tensor]
+                  ) -> "DistArray":
+        """Apply a row-count-preserving function to the local shard
+        (map_blocks analog, dask/array/core.py:459)."""
+        return self._like(fn(self.local))
+
+    def astype(self, dtype: torch.dtype) -> "DistArray":
+        return self._like(self.local.to(dtype))
+
+    def copy(self) -> "DistArray":
+        return self._like(self.local.clone())
+
+    def blocks(self, block_rows: Optional[int] = None
+               ) -> Iterator[torch.Tensor]:
+        """Iterate the local shard in row blocks (partial_fit slabs)."""
+        if block_rows is None:
+            block_rows = int(config.get("array.block_rows"))
+        m = self.local.shape[0]
+        for lo in range(0, m, block_rows):
+            yield self.local[lo:min(m, lo + block_rows)]
+
+    # --------------------------------------------------------- reductions
+    def _as2d(self) -> torch.Tensor:
+        return self.local if self.local.dim() == 2 else self.local[:, None]
+
+    def _vec(self, t: torch.Tensor) -> torch.Tensor:
+        return t[0] if self.local.dim() == 1 else t
+
+    def moments(self) -> Tuple[float, torch.Tensor, torch.Tensor]:
+        """Global (n, per-column mean f64, per-column M2 f64) — exact
+        Chan merge of per-rank one-pass partials (C1)."""
+        n_l, s_l, m2_l = ops.col_moments(self._as2d())
+        if self.comm.world_size > 1:
+            parts = self.comm.all_gather(
+                torch.cat([n_l[None], s_l, m2_l]).contiguous())
+            f = s_l.shape[0]
+            ns = torch.stack([p[0] for p in parts])
+            sums = torch.stack([p[1:1 + f] for p in parts])
+            m2s = torch.stack([p[1 + f:] for p in parts])
+            n, s, m2 = ops.merge_moments(ns, sums, m2s)
+        else:
+            n, s, m2 = n_l, s_l, m2_l
+        nn = float(n)
+        mean = s / max(nn, 1.0)
+        return nn, self._vec(mean), self._vec(m2)
+
+    def mean(self, axis: int = 0) -> torch.Tensor:
+        assert axis == 0
+        _, mean, _ = self.moments()
+        return mean
+
+    def var(self, axis: int = 0, ddof: int = 0) -> torch.Tensor:
+        assert axis == 0
+        n, _, m2 = self.moments()
+        denom = n - ddof
+        if denom <= 0:
+            return torch.full_like(m2, float("nan"))
+        return m2 / denom
+
+    def std(self, axis: int = 0, ddof: int = 0) -> torch.Tensor:
+        return self.var(axis, ddof).sqrt()
+
+    def sum(self, axis: int = 0) -> torch.Tensor:
+        assert axis == 0
+        s = ops.col_sum(self._as2d())
+        if self.comm.world_size > 1:
+            self.comm.all_reduce(s)
+        return self._vec(s)
+
+    def min(self, axis: int = 0) -> torch.Tensor:
+        assert axis == 0
+        mn, _ = ops.col_minmax(self._as2d())
+        if self.comm.world_size > 1:
+            self.comm.all_reduce(mn, op="min")
+        return self._vec(mn)
+
+    def max(self, axis: int = 0) -> torch.Tensor:
+        assert axis == 0
+        _, mx = ops.col_minmax(self._as2d())
+        if self.comm.world_size > 1:
+            self.comm.all_reduce(mx, op="max")
+        return self._vec(mx)
+
+    # ------------------------------------------------------------- linalg
+    def dot(self, w: torch.Tensor) -> "DistArray":
+        """X·w with replicated w — purely local per shard."""
+        return DistArray(self._as2d() @ w.to(self.local.dtype),
+                         self.global_rows, self.row_offset, self.comm)
+
+    def xt_dot(self, v: "DistArray", dtype: torch.dtype = torch.float64
+               ) -> torch.Tensor:
+        """Xᵀ·v — per-shard GEMM + one allreduce (C2), replicated result."""
+        g = (self._as2d().to(dtype).t() @ v._as2d().to(dtype))
+        if self.comm.world_size > 1:
+            self.comm.all_reduce(g)
+        return g
+
+    def gram(self, dtype: torch.dtype = torch.float64) -> torch.Tensor:
+        """XᵀX — per-shard syrk + allreduce (Hessian / covariance path)."""
+        x = self._as2d().to(dtype)
+        g = x.t() @ x
+        if self.comm.world_size > 1:
+            self.comm.all_reduce(g)
+        return g
+
+    # ------------------------------------------------------------ gather
+    def gather(self) -> torch.Tensor:
+        """Materialize the full array on every rank (tests / small results)."""
+        if self.comm.world_size == 1:
+            return self.local
+        parts = self.comm.all_gather_varlen(self.local.contiguous())
+        return torch.cat(parts, dim=0)
+
+    def to_numpy(self) -> np.ndarray:
+        return self.gather().cpu().numpy()
+
+    def persist(self) -> "DistArray":
+        """Eager framework: already resident. Kept for API parity."""
+        return self
+
+
+# ------------------------------------------------------------------ factories
+def shard_rows(full: torch.Tensor, comm: Optional[Communicator] = None
+               ) -> DistArray:
+    """Every rank holds/sees the same full tensor; keep only this rank's rows."""
+    comm = comm if comm is not None else comm_mod.default()
+    lo, hi = comm.shard_bounds(full.shape[0])
+    local = full[lo:hi].to(comm.device).contiguous()
+    return DistArray(local, full.shape[0], lo, comm)
+
+
+def from_numpy(a: np.ndarray, comm: Optional[Communicator] = None
+               ) -> DistArray:
+    return shard_rows(torch.from_numpy(np.ascontiguousarray(a)), comm)
+
+
+def from_local(local: torch.Tensor, comm: Optional[Communicator] = None
+               ) -> DistArray:
+    """Each rank provides its own shard; offsets/global size are derived."""
+    comm = comm if comm is not None else comm_mod.default()
+    if comm.world_size == 1:
+        return DistArray(local, local.shape[0], 0, comm)
+    n = torch.tensor([local.shape[0]], dtype=torch.int64,
+                     device=comm.device if comm.device.type == "cuda" else "cpu")
+    counts = [int(c.item()) for c in comm.all_gather(n)]
+    offset = sum(counts[: comm.rank])
+    return DistArray(local, sum(counts), offset, comm)

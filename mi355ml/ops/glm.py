@@ -1,0 +1,67 @@
+"""Fused GLM objective+gradient op (K1+K2+K3 in SURVEY §2.7).
+
+The reference evaluates a GLM objective as three separate dask graphs per
+lbfgs iteration — Xβ tensordot, pointwise family ops, Xᵀr tensordot + sum
+tree — i.e. ≥2 full passes over X plus graph overhead. Here the
+whole objective callback is ONE fused HIP kernel pass over the shard
+(X staged once through LDS, used for both z=X·w and the g=Xᵀr partial), so
+the per-iteration cost is bytes(X)/HBM-bandwidth.
+
+Families (dask_glm/families.py parity): logistic, normal, poisson.
+  logistic: loss=softplus(z)-y·z            r=sigmoid(z)-y
+  normal:   loss=0.5·(z-y)²                 r=z-y
+  poisson:  loss=exp(z)-y·z                 r=exp(z)-y
+(z = x·w + b; loss summed over rows; grad_w = Σ r·x, grad_b = Σ r.)
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from ._ext import hip_ext, use_hip
+
+FAMILIES = ("logistic", "normal", "poisson")
+_FAMILY_ID = {f: i for i, f in enumerate(FAMILIES)}
+
+
+def glm_value_grad(x: torch.Tensor, y: torch.Tensor, w: torch.Tensor,
+                   b: torch.Tensor, family: str = "logistic",
+                   ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Per-shard (loss_sum, grad_w, grad_b), all float64.
+
+    x: [m, d] f32; y: [m] f32; w: [d] f32; b: scalar f32 tensor.
+    """
+    assert family in _FAMILY_ID, family
+    assert x.dim() == 2 and y.dim() == 1 and w.dim() == 1
+    if use_hip(x):
+        ext = hip_ext()
+        loss, gw, gb = ext.glm_value_grad(
+            x.contiguous(), y.contiguous(), w.contiguous(),
+            float(b), _FAMILY_ID[family])
+        return loss, gw, gb
+    xd = x.to(torch.float64)
+    yd = y.to(torch.float64)
+    wd = w.to(torch.float64)
+    z = xd @ wd + float(b)
+    if family == "logistic":
+        loss = (torch.nn.functional.softplus(z) - yd * z).sum()
+        r = torch.sigmoid(z) - yd
+    elif family == "normal":
+        diff = z - yd
+        loss = 0.5 * (diff * diff).sum()
+        r = diff
+    else:  # poisson
+        ez = torch.exp(z)
+        loss = (ez - yd * z).sum()
+        r = ez - yd
+    gw = xd.t() @ r
+    gb = r.sum()
+    return loss, gw, gb
+
+
+def glm_predict_linear(x: torch.Tensor, w: torch.Tensor, b: float
+                       ) -> torch.Tensor:
+    """z = X·w + b for a shard (library GEMV; not a hot op)."""
+    return x @ w.to(x.dtype) + b

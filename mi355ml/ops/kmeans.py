@@ -1,0 +1,79 @@
+"""Fused KMeans assignment + centroid-partial op (K4+K5 in SURVEY §2.7).
+
+The reference's Lloyd iteration runs pairwise_distances_argmin_min as a
+map_blocks task (centers broadcast into every task) followed by per-block
+centroid sums and a tree-reduce. Here it is one fused HIP
+kernel per shard: the x·Cᵀ tile GEMM runs on f32 MFMA, the argmin rides in
+the epilogue, and per-workgroup centroid sums/counts land in a slab that is
+tree-reduced on-device; the cross-rank finish is one ncclAllReduce (C3).
+
+dist²(x, c) = ‖x‖² − 2·x·c + ‖c‖², with the ‖x‖² term dropped for argmin
+and added back only for inertia.
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from ._ext import hip_ext, use_hip
+
+
+def kmeans_assign(x: torch.Tensor, centers: torch.Tensor
+                  ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Per-shard labels + centroid partials.
+
+    Returns (labels int32 [m], sums f64 [k, d], counts f64 [k],
+    inertia f64 scalar). Empty-shard safe.
+    """
+    assert x.dim() == 2 and centers.dim() == 2
+    k, d = centers.shape
+    if x.shape[0] == 0:
+        return (torch.empty(0, dtype=torch.int32, device=x.device),
+                torch.zeros(k, d, dtype=torch.float64, device=x.device),
+                torch.zeros(k, dtype=torch.float64, device=x.device),
+                torch.zeros((), dtype=torch.float64, device=x.device))
+    if use_hip(x):
+        ext = hip_ext()
+        labels, sums, counts, inertia = ext.kmeans_assign(
+            x.contiguous(), centers.contiguous())
+        return labels, sums, counts, inertia
+    labels, mindist = closest_center(x, centers)
+    sums = torch.zeros(k, d, dtype=torch.float64, device=x.device)
+    sums.index_add_(0, labels.long(), x.to(torch.float64))
+    counts = torch.bincount(labels.long(), minlength=k).to(torch.float64)
+    inertia = mindist.to(torch.float64).sum()
+    return labels, sums, counts, inertia
+
+
+def closest_center(x: torch.Tensor, centers: torch.Tensor
+                   ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(labels int32, squared distance f32) of each row to its closest center.
+
+    (pairwise_distances_argmin_min parity; dist clamped at 0.)
+    """
+    assert x.dim() == 2 and centers.dim() == 2
+    if x.shape[0] == 0:
+        return (torch.empty(0, dtype=torch.int32, device=x.device),
+                torch.empty(0, dtype=x.dtype, device=x.device))
+    if use_hip(x):
+        ext = hip_ext()
+        labels, dists = ext.closest_center(x.contiguous(), centers.contiguous())
+        return labels, dists
+    cn = (centers * centers).sum(dim=1)
+    # blocked to bound the [m, k] temporary
+    m = x.shape[0]
+    block = max(1, min(m, 8_388_608 // max(1, centers.shape[0])))
+    labels = torch.empty(m, dtype=torch.int32, device=x.device)
+    dists = torch.empty(m, dtype=x.dtype, device=x.device)
+    for lo in range(0, m, block):
+        hi = min(m, lo + block)
+        xb = x[lo:hi]
+        half = xb @ centers.t()
+        dd = cn[None, :] - 2.0 * half
+        md, lab = dd.min(dim=1)
+        xn = (xb * xb).sum(dim=1)
+        labels[lo:hi] = lab.to(torch.int32)
+        dists[lo:hi] = (md + xn).clamp_(min=0)
+    return labels, dists

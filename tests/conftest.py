@@ -1,0 +1,55 @@
+import os
+import socket
+
+import pytest
+import torch
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X GPU (run with -m gpu)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU in this environment")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _mp_entry(rank, world, port, fn, args):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import mi355ml.parallel.comm as comm
+    comm.init(backend="gloo")
+    try:
+        fn(rank, world, *args)
+    finally:
+        comm.shutdown()
+
+
+def run_multiprocess(fn, world: int = 2, args: tuple = ()):
+    """Run fn(rank, world, *args) on `world` gloo-connected processes
+    (the gen_cluster analog — SURVEY §4 tier 2)."""
+    import torch.multiprocessing as mp
+    port = free_port()
+    mp.start_processes(_mp_entry, args=(world, port, fn, args),
+                       nprocs=world, join=True, start_method="spawn")
+
+
+@pytest.fixture
+def mp_runner():
+    return run_multiprocess

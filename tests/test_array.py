@@ -1,0 +1,75 @@
+"""DistArray reductions vs numpy oracle, single- and multi-process."""
+
+import numpy as np
+import pytest
+import torch
+
+import mi355ml
+from mi355ml.core.array import from_numpy, shard_rows
+
+
+def _data(n=1000, f=7, seed=0):
+    rng = np.random.default_rng(seed)
+    return (rng.standard_normal((n, f)) * rng.uniform(0.5, 3, f)
+            + rng.uniform(-2, 2, f)).astype(np.float32)
+
+
+def test_moments_single():
+    a = _data()
+    X = from_numpy(a)
+    np.testing.assert_allclose(X.mean().numpy(), a.mean(0), rtol=1e-6,
+                               atol=1e-6)
+    np.testing.assert_allclose(X.var().numpy(), a.var(0), rtol=1e-5,
+                               atol=1e-6)
+    np.testing.assert_allclose(X.var(ddof=1).numpy(), a.var(0, ddof=1),
+                               rtol=1e-5, atol=1e-6)
+    np.testing.assert_allclose(X.sum().numpy(), a.sum(0, dtype=np.float64),
+                               rtol=1e-6)
+
+
+def test_minmax_gather():
+    a = _data()
+    X = from_numpy(a)
+    np.testing.assert_allclose(X.min().numpy(), a.min(0))
+    np.testing.assert_allclose(X.max().numpy(), a.max(0))
+    np.testing.assert_allclose(X.gather().numpy(), a)
+
+
+def test_dot_xtdot():
+    a = _data()
+    X = from_numpy(a)
+    w = np.random.default_rng(1).standard_normal(a.shape[1]).astype(np.float32)
+    z = X.dot(torch.from_numpy(w))
+    np.testing.assert_allclose(z.gather().numpy(), a @ w, rtol=1e-4,
+                               atol=1e-4)
+    v = from_numpy((a @ w).astype(np.float32))
+    g = X.xt_dot(v)
+    np.testing.assert_allclose(g.numpy()[:, 0],
+                               a.astype(np.float64).T @ (a @ w).astype(np.float64),
+                               rtol=1e-5)
+
+
+def test_1d_moments():
+    rng = np.random.default_rng(2)
+    a = rng.standard_normal(500).astype(np.float32)
+    y = from_numpy(a)
+    assert y.shape == (500,)
+    np.testing.assert_allclose(float(y.mean()), a.mean(), rtol=1e-6, atol=1e-6)
+    np.testing.assert_allclose(float(y.var()), a.var(), rtol=1e-5, atol=1e-6)
+
+
+def _dist_moments_body(rank, world, a_bytes, shape):
+    a = np.frombuffer(a_bytes, dtype=np.float32).reshape(shape).copy()
+    X = from_numpy(a)
+    assert X.local.shape[0] < a.shape[0]
+    np.testing.assert_allclose(X.mean().numpy(), a.mean(0), rtol=1e-6,
+                               atol=1e-6)
+    np.testing.assert_allclose(X.var().numpy(), a.var(0), rtol=1e-5,
+                               atol=1e-6)
+    np.testing.assert_allclose(X.min().numpy(), a.min(0))
+    np.testing.assert_allclose(X.gather().numpy(), a)
+
+
+def test_dist_moments_world2(mp_runner):
+    a = _data(501, 5, seed=3)  # uneven split
+    mp_runner(_dist_moments_body, world=2, args=(a.tobytes(), a.shape))

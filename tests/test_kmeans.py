@@ -1,0 +1,82 @@
+"""KMeans vs sklearn oracle (tolerance-based — SURVEY §4)."""
+
+import numpy as np
+import joblib
+from sklearn.cluster import KMeans as SkKMeans
+
+from mi355ml.core.array import from_numpy
+from mi355ml.cluster import KMeans
+
+
+def _blobs(n=4000, f=5, k=4, seed=0):
+    rng = np.random.default_rng(seed)
+    centers = rng.uniform(-10, 10, (k, f))
+    lab = rng.integers(0, k, n)
+    X = (centers[lab] + rng.standard_normal((n, f))).astype(np.float32)
+    return X, lab
+
+
+def test_kmeans_matches_sklearn_quality():
+    X, _ = _blobs()
+    ours = KMeans(n_clusters=4, random_state=0, max_iter=100).fit(
+        from_numpy(X))
+    ref = SkKMeans(n_clusters=4, n_init=10, random_state=0).fit(X)
+    assert ours.inertia_ <= ref.inertia_ * 1.05
+    assert ours.n_iter_ >= 1
+    labels = ours.labels_.gather().numpy()
+    assert labels.shape == (X.shape[0],)
+    assert len(np.unique(labels)) == 4
+
+
+def test_kmeans_random_init():
+    X, _ = _blobs(seed=1)
+    ours = KMeans(n_clusters=4, init="random", random_state=0).fit(
+        from_numpy(X))
+    ref = SkKMeans(n_clusters=4, n_init=10, random_state=0).fit(X)
+    assert ours.inertia_ <= ref.inertia_ * 1.25
+
+
+def test_kmeans_explicit_init_predict_transform():
+    X, _ = _blobs(seed=2)
+    init = X[:4].copy()
+    ours = KMeans(n_clusters=4, init=init, max_iter=50).fit(from_numpy(X))
+    pred = ours.predict(from_numpy(X)).gather().numpy()
+    np.testing.assert_array_equal(pred, ours.labels_.gather().numpy())
+    d = ours.transform(from_numpy(X)).gather().numpy()
+    assert d.shape == (X.shape[0], 4)
+    np.testing.assert_array_equal(d.argmin(1), pred)
+    # score = -inertia at the fitted centers
+    assert abs(-ours.score(from_numpy(X)) - ours.inertia_) / ours.inertia_ < 1e-3
+
+
+def test_kmeans_deterministic():
+    X, _ = _blobs(seed=3)
+    a = KMeans(n_clusters=4, random_state=42).fit(from_numpy(X))
+    b = KMeans(n_clusters=4, random_state=42).fit(from_numpy(X))
+    np.testing.assert_array_equal(a.cluster_centers_.numpy(),
+                                  b.cluster_centers_.numpy())
+
+
+def test_kmeans_joblib_roundtrip(tmp_path):
+    X, _ = _blobs(seed=4)
+    km = KMeans(n_clusters=4, random_state=0).fit(from_numpy(X))
+    p = tmp_path / "km.joblib"
+    joblib.dump(km, p)
+    km2 = joblib.load(p)
+    np.testing.assert_array_equal(km.cluster_centers_.numpy(),
+                                  km2.cluster_centers_.numpy())
+    pred = km2.predict(from_numpy(X)).gather().numpy()
+    np.testing.assert_array_equal(pred, km.labels_.gather().numpy())
+
+
+def _kmeans_world2(rank, world, X_bytes, n, f, ref_inertia):
+    X = np.frombuffer(X_bytes, dtype=np.float32).reshape(n, f).copy()
+    ours = KMeans(n_clusters=4, random_state=0).fit(from_numpy(X))
+    assert ours.inertia_ <= ref_inertia * 1.1
+
+
+def test_kmeans_world2(mp_runner):
+    X, _ = _blobs(n=2001, seed=5)
+    ref = SkKMeans(n_clusters=4, n_init=10, random_state=0).fit(X)
+    mp_runner(_kmeans_world2, world=2,
+              args=(X.tobytes(), X.shape[0], X.shape[1], ref.inertia_))
