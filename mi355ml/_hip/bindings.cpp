@@ -103,10 +103,11 @@ std::vector<torch::Tensor> kmeans_assign(torch::Tensor x,
   const int d = (int)x.size(1);
   const int k = (int)centers.size(0);
   TORCH_CHECK(centers.size(1) == d);
-  TORCH_CHECK(d <= 128, "fused kmeans kernel supports d <= 128 (got ", d,
-              "); use the rocBLAS-composed path for wider data");
+  TORCH_CHECK(k <= 256, "fused kmeans kernel supports k <= 256 (got ", k,
+              "); use the rocBLAS-composed path for more centers");
   TORCH_CHECK(kmeans_smem_bytes(k, d, 1) <= 160 * 1024 - 1024,
-              "k*d too large for LDS bins");
+              "k*d too large for LDS (", kmeans_smem_bytes(k, d, 1),
+              " bytes)");
   const int nb = (int)std::max(1L, std::min(512L, (m + 255) / 256));
   auto f32 = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
   auto f64 = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
@@ -132,7 +133,9 @@ std::vector<torch::Tensor> closest_center(torch::Tensor x,
   const int d = (int)x.size(1);
   const int k = (int)centers.size(0);
   TORCH_CHECK(centers.size(1) == d);
-  TORCH_CHECK(d <= 128, "fused closest_center supports d <= 128");
+  TORCH_CHECK(k <= 256, "fused closest_center supports k <= 256");
+  TORCH_CHECK(kmeans_smem_bytes(k, d, 0) <= 160 * 1024 - 1024,
+              "k*d too large for LDS");
   const int nb = (int)std::max(1L, std::min(2048L, (m + 255) / 256));
   auto labels = torch::empty({m}, torch::TensorOptions()
                                       .dtype(torch::kInt32)

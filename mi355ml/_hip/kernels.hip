@@ -154,10 +154,18 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
     const long r0 = tile * GLM_TR;
     const int rows = (int)min((long)GLM_TR, m - r0);
     __syncthreads();  // previous grad-phase done before overwriting xt
-    for (int i = tid; i < rows * d; i += NT) {
-      const int rr = i / d;
-      const int cc = i - rr * d;
-      xt[rr * ld + cc] = x[(r0 + rr) * (long)d + cc];
+    if (ld == d && (d & 3) == 0) {
+      // contiguous LDS image: float4 staging (G13 — vectorize)
+      const long total4 = ((long)rows * d) >> 2;
+      const float4* src4 = (const float4*)(x + r0 * (long)d);
+      float4* dst4 = (float4*)xt;
+      for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
+    } else {
+      for (int i = tid; i < rows * d; i += NT) {
+        const int rr = i / d;
+        const int cc = i - rr * d;
+        xt[rr * ld + cc] = x[(r0 + rr) * (long)d + cc];
+      }
     }
     for (int i = tid; i < rows; i += NT) yt[i] = y[r0 + i];
     __syncthreads();
@@ -231,78 +239,182 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
 }
 
 // ---------------------------------------------------------------------------
-// K4+K5: fused KMeans assignment (+ optional centroid partials).
-// One thread per row, x-row register-resident (d <= 128), centers staged in
-// LDS (lane-uniform reads broadcast). Centroid partial sums/counts
-// accumulate in per-WG LDS bins (LDS atomics) and flush once per WG into a
-// slab; inertia reduces per-WG in f64.
+// K4+K5: fused KMeans assignment (+ optional centroid partials), MFMA f32.
+//
+// The x·Cᵀ tile GEMM runs on v_mfma_f32_16x16x4_f32 (exact f32 at the
+// 157 TF vector peak — MI355X_MICROARCH §MFMA): each wave owns a 16-row
+// group of the 64-row LDS tile and accumulates all ⌈k/16⌉ center tiles as
+// independent MFMA chains (pipelines the 40-cycle dependent latency).
+// dist(x,c) = ‖x‖² − 2·x·c + ‖c‖²: accumulators start at −‖c‖²/2 so the
+// argmin becomes an argmax over acc = x·c − ‖c‖²/2 in the epilogue
+// (shfl_xor reduce over the 16 lanes that share a row group).
+// Centroid partials land in per-WG LDS bins (LDS atomics, labels staged in
+// LDS) and flush once per WG to a slab — no global atomics, deterministic.
 // ---------------------------------------------------------------------------
-#define KM_MAXD 128
+#define KM_TR 64  // rows per tile (4 waves x 16 rows)
 
-template <int UPDATE>
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// NTILES = exact compile-time center-tile count (k16/16): keeps the
+// accumulator array statically indexed AND the MFMA inner loop branch-free —
+// runtime-guarded unrolls made hipcc round-trip every accumulator through one
+// a[0:3] AGPR window (accvgpr_write/read + s_nop 9 per MFMA, fully serial).
+template <int UPDATE, int NTILES>
 __global__ __launch_bounds__(NT) void kmeans_assign_k(
     const float* __restrict__ x, const float* __restrict__ cen, long m,
     int d, int k, int* __restrict__ labels, float* __restrict__ dists,
     float* __restrict__ sums_slab, float* __restrict__ cnt_slab,
     double* __restrict__ inertia_slab) {
+  const int d4 = (d + 3) & ~3;        // K padded to a multiple of 4
+  const int k16 = NTILES * 16;        // centers padded to a multiple of 16
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* sc = (float*)smem_raw;        // centers [k*d]
-  float* cn = sc + k * d;              // center norms [k]
-  float* bins = cn + ((k + 3) & ~3);   // UPDATE: [k*d]
-  float* bcnt = bins + (UPDATE ? k * d : 0);  // UPDATE: [k]
-  double* red = (double*)(smem_raw +
-      (((k * d + ((k + 3) & ~3) + (UPDATE ? (k * d + ((k + 3) & ~3)) : 0)) *
-        sizeof(float) + 15) & ~15));   // [NT]
+  float* ct = (float*)smem_raw;                  // centers [k16 * d4]
+  float* nhc = ct + k16 * d4;                    // -0.5*‖c‖² [k16]
+  float* xt = nhc + k16;                         // x tile [KM_TR * d4]
+  float* xn = xt + KM_TR * d4;                   // row norms [KM_TR]
+  int* rlab = (int*)(xn + KM_TR);                // row labels [KM_TR]
+  float* bins = (float*)(rlab + KM_TR);          // UPDATE: [k*d]
+  float* bcnt = bins + (UPDATE ? k * d : 0);     // UPDATE: [k]
+  double* red = (double*)(smem_raw + ((((size_t)(k16 * d4 + k16 +
+      KM_TR * d4 + 2 * KM_TR + (UPDATE ? (k * d + k) : 0)) * 4 + 15) & ~15)));
   const int tid = threadIdx.x;
-  for (int i = tid; i < k * d; i += NT) sc[i] = cen[i];
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int lj = lane & 15;   // fragment column (center within tile)
+  const int lk = lane >> 4;   // fragment K index (0..3)
+
+  // stage centers (zero-padded) + norms
+  for (int i = tid; i < k16 * d4; i += NT) ct[i] = 0.0f;
+  __syncthreads();
+  for (int i = tid; i < k * d; i += NT) {
+    const int j = i / d;
+    ct[j * d4 + (i - j * d)] = cen[i];
+  }
   if (UPDATE) {
     for (int i = tid; i < k * d; i += NT) bins[i] = 0.0f;
     for (int i = tid; i < k; i += NT) bcnt[i] = 0.0f;
   }
   __syncthreads();
-  for (int c = tid; c < k; c += NT) {
-    float s = 0.0f;
-    const float* cp = sc + c * d;
-    for (int j = 0; j < d; ++j) s += cp[j] * cp[j];
-    cn[c] = s;
+  for (int j = tid; j < k16; j += NT) {
+    if (j < k) {
+      float s = 0.0f;
+      const float* cp = ct + j * d4;
+      for (int kk = 0; kk < d4; ++kk) s += cp[kk] * cp[kk];
+      nhc[j] = -0.5f * s;
+    } else {
+      nhc[j] = -INFINITY;  // padded centers never win the argmax
+    }
   }
   __syncthreads();
 
-  float xr[KM_MAXD];
   double iacc = 0.0;
-  long r = (long)blockIdx.x * NT + tid;
-  const long stride = (long)gridDim.x * NT;
-  for (; r < m; r += stride) {
-    const float* xrow = x + r * (long)d;
-    float xn = 0.0f;
-#pragma unroll 4
-    for (int j = 0; j < d; ++j) {
-      const float v = xrow[j];
-      xr[j] = v;
-      xn += v * v;
-    }
-    float best = INFINITY;
-    int bi = 0;
-    for (int c = 0; c < k; ++c) {
-      const float* cp = sc + c * d;
-      float dot = 0.0f;
-#pragma unroll 4
-      for (int j = 0; j < d; ++j) dot = fmaf(xr[j], cp[j], dot);
-      const float dist = cn[c] - 2.0f * dot;
-      if (dist < best) {
-        best = dist;
-        bi = c;
+  const long tiles = (m + KM_TR - 1) / KM_TR;
+  for (long tile = blockIdx.x; tile < tiles; tile += gridDim.x) {
+    const long r0 = tile * KM_TR;
+    const int rows = (int)min((long)KM_TR, m - r0);
+    __syncthreads();  // previous iteration fully done before restaging
+    // ---- stage x tile (float4 path when K needs no padding) ----
+    if (d == d4) {
+      const long total4 = ((long)rows * d) >> 2;
+      const float4* src4 = (const float4*)(x + r0 * (long)d);
+      float4* dst4 = (float4*)xt;
+      for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
+      const int tail = (rows * d) & 3;
+      for (int i = tid; i < tail; ++i)
+        xt[rows * d - tail + i] = x[r0 * (long)d + rows * d - tail + i];
+    } else {
+      for (int i = tid; i < rows * d; i += NT) {
+        const int rr = i / d;
+        xt[rr * d4 + (i - rr * d)] = x[r0 * (long)d + i];
+      }
+      for (int i = tid; i < rows * (d4 - d); i += NT) {
+        const int rr = i / (d4 - d);
+        xt[rr * d4 + d + (i - rr * (d4 - d))] = 0.0f;
       }
     }
-    labels[r] = bi;
-    const float d2 = fmaxf(best + xn, 0.0f);
+    __syncthreads();
+    // ---- row norms ----
+    for (int r = tid; r < rows; r += NT) {
+      float s = 0.0f;
+      const float* xp = xt + r * d4;
+      for (int j = 0; j < d4; ++j) s += xp[j] * xp[j];
+      xn[r] = s;
+    }
+    __syncthreads();
+    // ---- MFMA distance GEMM: this wave's 16 rows vs all center tiles ----
+    {
+      const int ri = wave * 16 + lj;  // A-fragment row for this lane
+      f32x4 acc[NTILES];  // statically indexed, branch-free unrolls
+#pragma unroll
+      for (int t = 0; t < NTILES; ++t) {
+        const float v = nhc[16 * t + lj];
+        acc[t] = (f32x4){v, v, v, v};
+      }
+      const float* xrow = xt + ri * d4 + lk;
+      const float* cb = ct + lj * d4 + lk;
+      for (int k0 = 0; k0 < d4; k0 += 4) {
+        const float a = xrow[k0];
+#pragma unroll
+        for (int t = 0; t < NTILES; ++t) {
+          const float b = cb[16 * t * d4 + k0];
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[t], 0, 0, 0);
+        }
+      }
+      // epilogue: per-lane running argmax over center tiles, then reduce
+      // across the 16 lanes of the row group (rows (lane>>4)*4 + reg).
+      float best0 = -INFINITY, best1 = -INFINITY, best2 = -INFINITY,
+            best3 = -INFINITY;
+      int bj0 = 0, bj1 = 0, bj2 = 0, bj3 = 0;
+#pragma unroll
+      for (int t = 0; t < NTILES; ++t) {
+        const int j = 16 * t + lj;
+        if (acc[t][0] > best0) { best0 = acc[t][0]; bj0 = j; }
+        if (acc[t][1] > best1) { best1 = acc[t][1]; bj1 = j; }
+        if (acc[t][2] > best2) { best2 = acc[t][2]; bj2 = j; }
+        if (acc[t][3] > best3) { best3 = acc[t][3]; bj3 = j; }
+      }
+#define KM_RED(BEST, BJ)                                                   \
+  for (int off = 8; off > 0; off >>= 1) {                                  \
+    const float ov = __shfl_xor(BEST, off, 16);                            \
+    const int oj = __shfl_xor(BJ, off, 16);                                \
+    if (ov > BEST || (ov == BEST && oj < BJ)) { BEST = ov; BJ = oj; }      \
+  }
+      KM_RED(best0, bj0)
+      KM_RED(best1, bj1)
+      KM_RED(best2, bj2)
+      KM_RED(best3, bj3)
+#undef KM_RED
+      if (lj == 0) {
+        const int rbase = wave * 16 + lk * 4;
+        const float bests[4] = {best0, best1, best2, best3};
+        const int bjs[4] = {bj0, bj1, bj2, bj3};
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          const int r = rbase + g;
+          if (r < rows) {
+            const float d2 = fmaxf(xn[r] - 2.0f * bests[g], 0.0f);
+            labels[r0 + r] = bjs[g];
+            if (UPDATE) {
+              rlab[r] = bjs[g];
+              iacc += (double)d2;
+            } else {
+              dists[r0 + r] = d2;
+            }
+          }
+        }
+      }
+    }
     if (UPDATE) {
-      iacc += (double)d2;
-      float* bp = bins + bi * d;
-      for (int j = 0; j < d; ++j) atomicAdd(&bp[j], xr[j]);
-      atomicAdd(&bcnt[bi], 1.0f);
-    } else {
-      dists[r] = d2;
+      __syncthreads();  // rlab ready
+      const int r = tid >> 2;
+      const int q = tid & 3;
+      if (r < rows) {
+        const int lr = rlab[r];
+        float* bp = bins + lr * d;
+        const float* xp = xt + r * d4;
+        for (int j = q; j < d; j += 4) atomicAdd(&bp[j], xp[j]);
+        if (q == 0) atomicAdd(&bcnt[lr], 1.0f);
+      }
     }
   }
 
@@ -357,6 +469,37 @@ extern "C" __global__ __launch_bounds__(NT) void affine_transform_k(
 // ---------------------------------------------------------------------------
 // C-visible launchers
 // ---------------------------------------------------------------------------
+static size_t kmeans_smem_bytes_impl(int k, int d, int update) {
+  const int d4 = (d + 3) & ~3;
+  const int k16 = (k + 15) & ~15;
+  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)64 * d4 + 2 * 64;
+  if (update) f32s += (size_t)k * d + k;
+  return (((f32s * 4 + 15) & ~15)) + NT * sizeof(double);
+}
+
+template <int UPDATE>
+static void kmeans_dispatch(const float* x, const float* cen, long m, int d,
+                            int k, int* labels, float* dists,
+                            float* sums_slab, float* cnt_slab,
+                            double* inertia_slab, int nblocks,
+                            hipStream_t stream) {
+  const size_t smem = kmeans_smem_bytes_impl(k, d, UPDATE);
+  const int ntiles = ((k + 15) & ~15) / 16;
+#define KM_LAUNCH(NTL)                                                      \
+  case NTL:                                                                 \
+    hipLaunchKernelGGL((kmeans_assign_k<UPDATE, NTL>), dim3(nblocks),       \
+                       dim3(NT), smem, stream, x, cen, m, d, k, labels,     \
+                       dists, sums_slab, cnt_slab, inertia_slab);           \
+    break
+  switch (ntiles) {
+    KM_LAUNCH(1); KM_LAUNCH(2); KM_LAUNCH(3); KM_LAUNCH(4);
+    KM_LAUNCH(5); KM_LAUNCH(6); KM_LAUNCH(7); KM_LAUNCH(8);
+    KM_LAUNCH(9); KM_LAUNCH(10); KM_LAUNCH(11); KM_LAUNCH(12);
+    KM_LAUNCH(13); KM_LAUNCH(14); KM_LAUNCH(15); KM_LAUNCH(16);
+  }
+#undef KM_LAUNCH
+}
+
 extern "C" {
 
 void launch_col_moments(const float* x, long m, int f, double* ns,
@@ -376,7 +519,9 @@ void launch_glm_value_grad(const float* x, const float* y, const float* w,
                            float b, long m, int d, int family,
                            double* loss_slab, double* gw_slab,
                            double* gb_slab, int nblocks, hipStream_t stream) {
-  const int ld = d | 1;  // odd LDS stride -> conflict-free row & col reads
+  // d%4==0: contiguous LDS image (float4 staging, <=2-way bank conflicts);
+  // otherwise odd stride (conflict-free, scalar staging).
+  const int ld = ((d & 3) == 0) ? d : (d | 1);
   const size_t smem =
       ((((((d + 3) & ~3) + GLM_TR * ld + 2 * GLM_TR) * sizeof(float) + 15) &
         ~15)) + 2 * NT * sizeof(double);
@@ -399,28 +544,22 @@ void launch_glm_value_grad(const float* x, const float* y, const float* w,
 }
 
 size_t kmeans_smem_bytes(int k, int d, int update) {
-  size_t f32s = (size_t)k * d + ((k + 3) & ~3);
-  if (update) f32s += (size_t)k * d + ((k + 3) & ~3);
-  return (((f32s * sizeof(float) + 15) & ~15)) + NT * sizeof(double);
+  return kmeans_smem_bytes_impl(k, d, update);
 }
 
 void launch_kmeans_assign(const float* x, const float* cen, long m, int d,
                           int k, int* labels, float* sums_slab,
                           float* cnt_slab, double* inertia_slab, int nblocks,
                           hipStream_t stream) {
-  const size_t smem = kmeans_smem_bytes(k, d, 1);
-  hipLaunchKernelGGL((kmeans_assign_k<1>), dim3(nblocks), dim3(NT), smem,
-                     stream, x, cen, m, d, k, labels, (float*)nullptr,
-                     sums_slab, cnt_slab, inertia_slab);
+  kmeans_dispatch<1>(x, cen, m, d, k, labels, nullptr, sums_slab, cnt_slab,
+                     inertia_slab, nblocks, stream);
 }
 
 void launch_closest_center(const float* x, const float* cen, long m, int d,
                            int k, int* labels, float* dists, int nblocks,
                            hipStream_t stream) {
-  const size_t smem = kmeans_smem_bytes(k, d, 0);
-  hipLaunchKernelGGL((kmeans_assign_k<0>), dim3(nblocks), dim3(NT), smem,
-                     stream, x, cen, m, d, k, labels, dists, (float*)nullptr,
-                     (float*)nullptr, (double*)nullptr);
+  kmeans_dispatch<0>(x, cen, m, d, k, labels, dists, nullptr, nullptr,
+                     nullptr, nblocks, stream);
 }
 
 void launch_affine_transform(const float* x, const float* a, const float* b,

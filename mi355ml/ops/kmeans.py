@@ -20,6 +20,22 @@ import torch
 from ._ext import hip_ext, use_hip
 
 
+def _fits_fused(k: int, d: int, update: bool) -> bool:
+    """Mirror of kernels.hip kmeans_smem_bytes: the fused kernel stages
+    centers + a 64-row x tile (+ bins) in the 160 KB LDS. Outside this
+    envelope the distance GEMM goes through rocBLAS (plain library GEMM,
+    sanctioned) + torch argmin/index_add — an explicit composed path, not a
+    silent eager fallback."""
+    if k > 256:
+        return False
+    d4 = (d + 3) & ~3
+    k16 = (k + 15) & ~15
+    f32s = k16 * d4 + k16 + 64 * d4 + 2 * 64
+    if update:
+        f32s += k * d + k
+    return f32s * 4 + 16 + 256 * 8 <= 160 * 1024 - 1024
+
+
 def kmeans_assign(x: torch.Tensor, centers: torch.Tensor
                   ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
     """Per-shard labels + centroid partials.
@@ -34,7 +50,7 @@ def kmeans_assign(x: torch.Tensor, centers: torch.Tensor
                 torch.zeros(k, d, dtype=torch.float64, device=x.device),
                 torch.zeros(k, dtype=torch.float64, device=x.device),
                 torch.zeros((), dtype=torch.float64, device=x.device))
-    if use_hip(x):
+    if use_hip(x) and _fits_fused(k, d, update=True):
         ext = hip_ext()
         labels, sums, counts, inertia = ext.kmeans_assign(
             x.contiguous(), centers.contiguous())
@@ -57,7 +73,7 @@ def closest_center(x: torch.Tensor, centers: torch.Tensor
     if x.shape[0] == 0:
         return (torch.empty(0, dtype=torch.int32, device=x.device),
                 torch.empty(0, dtype=x.dtype, device=x.device))
-    if use_hip(x):
+    if use_hip(x) and _fits_fused(centers.shape[0], x.shape[1], update=False):
         ext = hip_ext()
         labels, dists = ext.closest_center(x.contiguous(), centers.contiguous())
         return labels, dists
