@@ -251,7 +251,8 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
 // Centroid partials land in per-WG LDS bins (LDS atomics, labels staged in
 // LDS) and flush once per WG to a slab — no global atomics, deterministic.
 // ---------------------------------------------------------------------------
-#define KM_TR 64  // rows per tile (4 waves x 16 rows)
+#define KM_TR 64   // rows per tile (4 waves x 16 rows)
+#define KM_STG 8   // float4 staging registers per thread (d4 <= 128)
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
@@ -309,19 +310,41 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
 
   double iacc = 0.0;
   const long tiles = (m + KM_TR - 1) / KM_TR;
-  for (long tile = blockIdx.x; tile < tiles; tile += gridDim.x) {
+  // T14 issue-early/write-late staging: the NEXT tile's global loads are
+  // issued into registers before this tile's compute and written to LDS
+  // after it, so HBM latency hides under the MFMA phase (1 WG/CU: no other
+  // workgroup provides TLP). KM_STG covers d4 <= 128 (64*128/4/256 = 8).
+  const bool pipelined = (d == d4) && (KM_TR * d4 / 4 <= KM_STG * NT);
+  float4 stg[KM_STG];
+  long tile = blockIdx.x;
+  if (pipelined && tile < tiles) {
+    const int rows0 = (int)min((long)KM_TR, m - tile * KM_TR);
+    const float4* src4 = (const float4*)(x + tile * KM_TR * (long)d);
+    const int total4 = (rows0 * d) >> 2;
+#pragma unroll
+    for (int j = 0; j < KM_STG; ++j) {
+      const int i = tid + j * NT;
+      if (i < total4) stg[j] = src4[i];
+    }
+  }
+  for (; tile < tiles; tile += gridDim.x) {
     const long r0 = tile * KM_TR;
     const int rows = (int)min((long)KM_TR, m - r0);
     __syncthreads();  // previous iteration fully done before restaging
-    // ---- stage x tile (float4 path when K needs no padding) ----
-    if (d == d4) {
+    if (pipelined) {
+      // write-late: this tile's registers -> LDS
+      const int total4 = (rows * d) >> 2;
+      float4* dst4 = (float4*)xt;
+#pragma unroll
+      for (int j = 0; j < KM_STG; ++j) {
+        const int i = tid + j * NT;
+        if (i < total4) dst4[i] = stg[j];
+      }
+    } else if (d == d4) {
       const long total4 = ((long)rows * d) >> 2;
       const float4* src4 = (const float4*)(x + r0 * (long)d);
       float4* dst4 = (float4*)xt;
       for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
-      const int tail = (rows * d) & 3;
-      for (int i = tid; i < tail; ++i)
-        xt[rows * d - tail + i] = x[r0 * (long)d + rows * d - tail + i];
     } else {
       for (int i = tid; i < rows * d; i += NT) {
         const int rr = i / d;
@@ -333,12 +356,30 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       }
     }
     __syncthreads();
-    // ---- row norms ----
-    for (int r = tid; r < rows; r += NT) {
+    // issue-early: next tile's loads go out before this tile's compute
+    if (pipelined && tile + gridDim.x < tiles) {
+      const long rn0 = (tile + gridDim.x) * KM_TR;
+      const int rowsn = (int)min((long)KM_TR, m - rn0);
+      const float4* src4 = (const float4*)(x + rn0 * (long)d);
+      const int total4 = (rowsn * d) >> 2;
+#pragma unroll
+      for (int j = 0; j < KM_STG; ++j) {
+        const int i = tid + j * NT;
+        if (i < total4) stg[j] = src4[i];
+      }
+    }
+    // ---- row norms (wave-parallel: 4 lanes per row) ----
+    {
+      const int r = tid >> 2;
+      const int q = tid & 3;
       float s = 0.0f;
-      const float* xp = xt + r * d4;
-      for (int j = 0; j < d4; ++j) s += xp[j] * xp[j];
-      xn[r] = s;
+      if (r < rows) {
+        const float* xp = xt + r * d4;
+        for (int j = q; j < d4; j += 4) s += xp[j] * xp[j];
+      }
+      s += __shfl_xor(s, 1, 4);
+      s += __shfl_xor(s, 2, 4);
+      if (r < rows && q == 0) xn[r] = s;
     }
     __syncthreads();
     // ---- MFMA distance GEMM: this wave's 16 rows vs all center tiles ----
@@ -352,13 +393,27 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       }
       const float* xrow = xt + ri * d4 + lk;
       const float* cb = ct + lj * d4 + lk;
-      for (int k0 = 0; k0 < d4; k0 += 4) {
-        const float a = xrow[k0];
+      // software pipeline: fragment loads for step k0+4 issue before the
+      // MFMAs of step k0 (one lgkmcnt wait per step instead of per MFMA)
+      float a_cur = xrow[0];
+      float b_cur[NTILES];
 #pragma unroll
-        for (int t = 0; t < NTILES; ++t) {
-          const float b = cb[16 * t * d4 + k0];
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[t], 0, 0, 0);
+      for (int t = 0; t < NTILES; ++t) b_cur[t] = cb[16 * t * d4];
+      for (int k0 = 0; k0 < d4; k0 += 4) {
+        float a_nxt = 0.0f;
+        float b_nxt[NTILES];
+        if (k0 + 4 < d4) {
+          a_nxt = xrow[k0 + 4];
+#pragma unroll
+          for (int t = 0; t < NTILES; ++t) b_nxt[t] = cb[16 * t * d4 + k0 + 4];
         }
+#pragma unroll
+        for (int t = 0; t < NTILES; ++t)
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a_cur, b_cur[t],
+                                                        acc[t], 0, 0, 0);
+        a_cur = a_nxt;
+#pragma unroll
+        for (int t = 0; t < NTILES; ++t) b_cur[t] = b_nxt[t];
       }
       // epilogue: per-lane running argmax over center tiles, then reduce
       // across the 16 lanes of the row group (rows (lane>>4)*4 + reg).
@@ -406,15 +461,19 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
     }
     if (UPDATE) {
       __syncthreads();  // rlab ready
-      const int r = tid >> 2;
-      const int q = tid & 3;
-      if (r < rows) {
-        const int lr = rlab[r];
-        float* bp = bins + lr * d;
-        const float* xp = xt + r * d4;
-        for (int j = q; j < d; j += 4) atomicAdd(&bp[j], xp[j]);
-        if (q == 0) atomicAdd(&bcnt[lr], 1.0f);
+      // Column-partitioned accumulation: wave w owns columns
+      // [w*dq, (w+1)*dq); each column has exactly ONE owner lane, so plain
+      // LDS read-modify-write suffices (the per-element atomicAdd version
+      // serialized 64 scattered lanes per instruction — it was ~46% of the
+      // kernel's LDS-array time).
+      const int dq = (d + 3) >> 2;  // columns per wave
+      const int c = wave * dq + lane;
+      if (lane < dq && c < d) {
+        for (int r = 0; r < rows; ++r) {
+          bins[rlab[r] * d + c] += xt[r * d4 + c];
+        }
       }
+      if (tid < rows) atomicAdd(&bcnt[rlab[tid]], 1.0f);
     }
   }
 
