@@ -20,6 +20,8 @@ void launch_closest_center(const float*, const float*, long, int, int, int*,
                            float*, int, hipStream_t);
 void launch_affine_transform(const float*, const float*, const float*,
                              float*, long, int, int, hipStream_t);
+void launch_kmeans_probe(const float*, const float*, long, int, int, int*,
+                         float*, float*, double*, int, int, hipStream_t);
 size_t kmeans_smem_bytes(int k, int d, int update);
 }
 
@@ -149,6 +151,28 @@ std::vector<torch::Tensor> closest_center(torch::Tensor x,
   return {labels, dists};
 }
 
+torch::Tensor kmeans_phase_probe(torch::Tensor x, torch::Tensor centers,
+                                 long phases) {
+  const long m = x.size(0);
+  const int d = (int)x.size(1);
+  const int k = (int)centers.size(0);
+  const int nb = (int)std::max(1L, std::min(512L, (m + 255) / 256));
+  auto f32 = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
+  auto f64 = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
+  auto labels = torch::zeros({m}, torch::TensorOptions()
+                                      .dtype(torch::kInt32)
+                                      .device(x.device()));
+  auto sums_slab = torch::zeros({nb, k, d}, f32);
+  auto cnt_slab = torch::zeros({nb, k}, f32);
+  auto inertia_slab = torch::zeros({nb}, f64);
+  launch_kmeans_probe(x.data_ptr<float>(), centers.data_ptr<float>(), m, d,
+                      k, labels.data_ptr<int>(), sums_slab.data_ptr<float>(),
+                      cnt_slab.data_ptr<float>(),
+                      inertia_slab.data_ptr<double>(), nb, (int)phases,
+                      cur_stream());
+  return labels;
+}
+
 torch::Tensor affine_transform(torch::Tensor x, torch::Tensor a,
                                torch::Tensor b, torch::Tensor out) {
   check_f32_2d(x, "x");
@@ -179,4 +203,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused argmin-distance labels + squared distances");
   mod.def("affine_transform", &affine_transform,
           "fused per-column affine transform");
+  mod.def("kmeans_phase_probe", &kmeans_phase_probe,
+          "phase-mask ablation probe (bit0 stage, bit1 compute, bit2 update)");
 }

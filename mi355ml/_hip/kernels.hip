@@ -119,6 +119,38 @@ extern "C" __global__ __launch_bounds__(NT) void col_minmax_partial_k(
   }
 }
 
+// raw barrier: waits only the LDS queue, NOT vmcnt — a plain __syncthreads()
+// with an LDS-DMA (global_load_lds) in flight emits s_waitcnt vmcnt(0) and
+// drains the prefetch (cdna_hip_programming.md §5 'Pipelining across
+// barriers').
+__device__ __forceinline__ void lds_barrier() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+}
+
+__device__ __forceinline__ void vm_drain() {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+}
+
+// LDS-DMA one KM_TR x d4 tile (KB-exact when d%4==0): wave w copies chunks
+// w, w+4, ... of 1 KiB each (64 lanes x 16 B), dest lane-linear.
+__device__ __forceinline__ void glds_tile(const float* __restrict__ src,
+                                          float* dst, int nfloats, int wave,
+                                          int lane) {
+  auto gsrc = (const __attribute__((address_space(1))) char*)(const void*)src;
+  auto ldst = (__attribute__((address_space(3))) char*)(void*)dst;
+  const int nchunks = nfloats >> 8;  // 256 floats = 1 KiB per chunk
+  for (int ch = wave; ch < nchunks; ch += 4) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(gsrc + ch * 1024 +
+                                                        lane * 16),
+        (__attribute__((address_space(3))) void*)(ldst + ch * 1024), 16, 0, 0);
+  }
+}
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+
 // ---------------------------------------------------------------------------
 // K1+K2+K3: fused GLM objective + gradient.
 // One pass over X per objective evaluation: each 64-row tile is staged once
@@ -133,42 +165,83 @@ extern "C" __global__ __launch_bounds__(NT) void col_minmax_partial_k(
 template <int FAM>
 __global__ __launch_bounds__(NT) void glm_value_grad_k(
     const float* __restrict__ x, const float* __restrict__ y,
-    const float* __restrict__ w, float b, long m, int d, int ld,
+    const float* __restrict__ w, float b, long m, int d, int ld, int nbuf,
     double* __restrict__ loss_slab, double* __restrict__ gw_slab,
     double* __restrict__ gb_slab) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* sw = (float*)smem_raw;                    // [d]
-  float* xt = sw + ((d + 3) & ~3);                 // [TR * ld]
-  float* yt = xt + GLM_TR * ld;                    // [TR]
-  float* rres = yt + GLM_TR;                       // [TR]
+  float* xtbuf = sw + ((d + 3) & ~3);              // nbuf x [TR * ld]
+  float* ytbuf = xtbuf + nbuf * GLM_TR * ld;       // nbuf x [TR]
+  float* rres = ytbuf + nbuf * GLM_TR;             // [TR]
   double* red = (double*)(smem_raw +
-      (((((d + 3) & ~3) + GLM_TR * ld + 2 * GLM_TR) * sizeof(float) + 15) & ~15));
+      (((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
+        sizeof(float) + 15) & ~15));
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
   for (int i = tid; i < d; i += NT) sw[i] = w[i];
 
   double lacc = 0.0, gbacc = 0.0;
   double gacc0 = 0.0, gacc1 = 0.0;  // column slots tid and tid+NT
 
   const long tiles = (m + GLM_TR - 1) / GLM_TR;
-  for (long tile = blockIdx.x; tile < tiles; tile += gridDim.x) {
-    const long r0 = tile * GLM_TR;
-    const int rows = (int)min((long)GLM_TR, m - r0);
-    __syncthreads();  // previous grad-phase done before overwriting xt
+  // Double-buffered LDS-DMA staging (same pattern as the kmeans kernel):
+  // tile t+1 streams into buffer cur^1 by global_load_lds while tile t is
+  // consumed; raw barriers so the DMA stays in flight across them.
+  const bool use_glds = ((d & 3) == 0) && (ld == d);
+  const bool dbuf = (nbuf == 2);
+  int cur = 0;
+  long tile = blockIdx.x;
+
+  auto stage_direct = [&](float* xdst, float* ydst, long t) {
+    const long dr0 = t * GLM_TR;
+    const int drows = (int)min((long)GLM_TR, m - dr0);
     if (ld == d && (d & 3) == 0) {
-      // contiguous LDS image: float4 staging (G13 — vectorize)
-      const long total4 = ((long)rows * d) >> 2;
-      const float4* src4 = (const float4*)(x + r0 * (long)d);
-      float4* dst4 = (float4*)xt;
+      const long total4 = ((long)drows * d) >> 2;
+      const float4* src4 = (const float4*)(x + dr0 * (long)d);
+      float4* dst4 = (float4*)xdst;
       for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
     } else {
-      for (int i = tid; i < rows * d; i += NT) {
+      for (int i = tid; i < drows * d; i += NT) {
         const int rr = i / d;
-        const int cc = i - rr * d;
-        xt[rr * ld + cc] = x[(r0 + rr) * (long)d + cc];
+        xdst[rr * ld + (i - rr * d)] = x[dr0 * (long)d + i];
       }
     }
-    for (int i = tid; i < rows; i += NT) yt[i] = y[r0 + i];
-    __syncthreads();
+    for (int i = tid; i < drows; i += NT) ydst[i] = y[dr0 + i];
+  };
+
+  auto stage_glds = [&](float* xdst, float* ydst, long t) {
+    glds_tile(x + t * GLM_TR * (long)d, xdst, GLM_TR * d, wave, lane);
+    if (wave == 0) {  // y tile: one 4-byte-per-lane DMA (64 floats)
+      auto gsrc = (const __attribute__((address_space(1))) char*)
+          (const void*)(y + t * GLM_TR);
+      auto ldst = (__attribute__((address_space(3))) char*)(void*)ydst;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(gsrc + lane * 4),
+          (__attribute__((address_space(3))) void*)ldst, 4, 0, 0);
+    }
+  };
+
+  if (tile < tiles) {
+    if (use_glds && m - tile * GLM_TR >= GLM_TR) {
+      stage_glds(xtbuf, ytbuf, tile);
+      vm_drain();
+    } else {
+      stage_direct(xtbuf, ytbuf, tile);
+    }
+  }
+  lds_barrier();
+  for (; tile < tiles; tile += gridDim.x) {
+    const long r0 = tile * GLM_TR;
+    const int rows = (int)min((long)GLM_TR, m - r0);
+    float* xt = xtbuf + cur * GLM_TR * ld;
+    float* yt = ytbuf + cur * GLM_TR;
+    const long ntile = tile + gridDim.x;
+    const bool pf = dbuf && use_glds && ntile < tiles &&
+                    (m - ntile * GLM_TR) >= GLM_TR;
+    const int nxt = dbuf ? (cur ^ 1) : 0;
+    if (pf)
+      stage_glds(xtbuf + nxt * GLM_TR * ld, ytbuf + nxt * GLM_TR, ntile);
     // z-phase: 4 lanes per row
     {
       const int r = tid >> 2;
@@ -202,7 +275,7 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
         rres[r] = (float)ri;
       }
     }
-    __syncthreads();
+    lds_barrier();
     // grad-phase: thread owns columns tid and tid+NT
     if (tid < d) {
       double acc = 0.0;
@@ -218,6 +291,13 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
         acc += (double)rres[rr] * (double)xp[rr * ld];
       gacc1 += acc;
     }
+    if (pf) {
+      vm_drain();  // next tile's DMA has the whole compute to land
+    } else if (ntile < tiles) {
+      stage_direct(xtbuf + nxt * GLM_TR * ld, ytbuf + nxt * GLM_TR, ntile);
+    }
+    lds_barrier();
+    cur ^= dbuf ? 1 : 0;
   }
   // block reduction of loss / gb
   red[tid] = lacc;
@@ -252,15 +332,12 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
 // LDS) and flush once per WG to a slab — no global atomics, deterministic.
 // ---------------------------------------------------------------------------
 #define KM_TR 64   // rows per tile (4 waves x 16 rows)
-#define KM_STG 8   // float4 staging registers per thread (d4 <= 128)
-
-typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 // NTILES = exact compile-time center-tile count (k16/16): keeps the
 // accumulator array statically indexed AND the MFMA inner loop branch-free —
 // runtime-guarded unrolls made hipcc round-trip every accumulator through one
 // a[0:3] AGPR window (accvgpr_write/read + s_nop 9 per MFMA, fully serial).
-template <int UPDATE, int NTILES>
+template <int UPDATE, int NTILES, int PHASES = 7>
 __global__ __launch_bounds__(NT) void kmeans_assign_k(
     const float* __restrict__ x, const float* __restrict__ cen, long m,
     int d, int k, int* __restrict__ labels, float* __restrict__ dists,
@@ -271,13 +348,13 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* ct = (float*)smem_raw;                  // centers [k16 * d4]
   float* nhc = ct + k16 * d4;                    // -0.5*‖c‖² [k16]
-  float* xt = nhc + k16;                         // x tile [KM_TR * d4]
-  float* xn = xt + KM_TR * d4;                   // row norms [KM_TR]
+  float* xtbuf = nhc + k16;                      // 2 x-tile buffers
+  float* xn = xtbuf + 2 * KM_TR * d4;            // row norms [KM_TR]
   int* rlab = (int*)(xn + KM_TR);                // row labels [KM_TR]
   float* bins = (float*)(rlab + KM_TR);          // UPDATE: [k*d]
   float* bcnt = bins + (UPDATE ? k * d : 0);     // UPDATE: [k]
   double* red = (double*)(smem_raw + ((((size_t)(k16 * d4 + k16 +
-      KM_TR * d4 + 2 * KM_TR + (UPDATE ? (k * d + k) : 0)) * 4 + 15) & ~15)));
+      2 * KM_TR * d4 + 2 * KM_TR + (UPDATE ? (k * d + k) : 0)) * 4 + 15) & ~15)));
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -310,66 +387,57 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
 
   double iacc = 0.0;
   const long tiles = (m + KM_TR - 1) / KM_TR;
-  // T14 issue-early/write-late staging: the NEXT tile's global loads are
-  // issued into registers before this tile's compute and written to LDS
-  // after it, so HBM latency hides under the MFMA phase (1 WG/CU: no other
-  // workgroup provides TLP). KM_STG covers d4 <= 128 (64*128/4/256 = 8).
-  const bool pipelined = (d == d4) && (KM_TR * d4 / 4 <= KM_STG * NT);
-  float4 stg[KM_STG];
+  // Double-buffered LDS-DMA staging: tile t+1 streams into buffer cur^1 by
+  // global_load_lds while tile t computes out of buffer cur; one vmcnt
+  // drain + raw barrier per tile (the §5 'Pipelining across barriers'
+  // pattern: __syncthreads would drain the DMA early).
+  const bool use_glds = (d == d4) && ((PHASES & 1) != 0);
+  int cur = 0;
   long tile = blockIdx.x;
-  if (pipelined && tile < tiles) {
-    const int rows0 = (int)min((long)KM_TR, m - tile * KM_TR);
-    const float4* src4 = (const float4*)(x + tile * KM_TR * (long)d);
-    const int total4 = (rows0 * d) >> 2;
-#pragma unroll
-    for (int j = 0; j < KM_STG; ++j) {
-      const int i = tid + j * NT;
-      if (i < total4) stg[j] = src4[i];
+
+  auto stage_direct = [&](float* dst, long t) {
+    const long dr0 = t * KM_TR;
+    const int drows = (int)min((long)KM_TR, m - dr0);
+    if (d == d4) {
+      const long total4 = ((long)drows * d) >> 2;
+      const float4* src4 = (const float4*)(x + dr0 * (long)d);
+      float4* dst4 = (float4*)dst;
+      for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
+    } else {
+      for (int i = tid; i < drows * d; i += NT) {
+        const int rr = i / d;
+        dst[rr * d4 + (i - rr * d)] = x[dr0 * (long)d + i];
+      }
+      for (int i = tid; i < drows * (d4 - d); i += NT) {
+        const int rr = i / (d4 - d);
+        dst[rr * d4 + d + (i - rr * (d4 - d))] = 0.0f;
+      }
+    }
+  };
+
+  if ((PHASES & 1) && tile < tiles) {
+    // prologue: fill buffer 0 with this WG's first tile
+    if (use_glds && m - tile * KM_TR >= KM_TR) {
+      glds_tile(x + tile * KM_TR * (long)d, xtbuf, KM_TR * d4, wave, lane);
+      vm_drain();
+    } else {
+      stage_direct(xtbuf, tile);
     }
   }
+  lds_barrier();
   for (; tile < tiles; tile += gridDim.x) {
     const long r0 = tile * KM_TR;
     const int rows = (int)min((long)KM_TR, m - r0);
-    __syncthreads();  // previous iteration fully done before restaging
-    if (pipelined) {
-      // write-late: this tile's registers -> LDS
-      const int total4 = (rows * d) >> 2;
-      float4* dst4 = (float4*)xt;
-#pragma unroll
-      for (int j = 0; j < KM_STG; ++j) {
-        const int i = tid + j * NT;
-        if (i < total4) dst4[i] = stg[j];
-      }
-    } else if (d == d4) {
-      const long total4 = ((long)rows * d) >> 2;
-      const float4* src4 = (const float4*)(x + r0 * (long)d);
-      float4* dst4 = (float4*)xt;
-      for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
-    } else {
-      for (int i = tid; i < rows * d; i += NT) {
-        const int rr = i / d;
-        xt[rr * d4 + (i - rr * d)] = x[r0 * (long)d + i];
-      }
-      for (int i = tid; i < rows * (d4 - d); i += NT) {
-        const int rr = i / (d4 - d);
-        xt[rr * d4 + d + (i - rr * (d4 - d))] = 0.0f;
-      }
-    }
-    __syncthreads();
-    // issue-early: next tile's loads go out before this tile's compute
-    if (pipelined && tile + gridDim.x < tiles) {
-      const long rn0 = (tile + gridDim.x) * KM_TR;
-      const int rowsn = (int)min((long)KM_TR, m - rn0);
-      const float4* src4 = (const float4*)(x + rn0 * (long)d);
-      const int total4 = (rowsn * d) >> 2;
-#pragma unroll
-      for (int j = 0; j < KM_STG; ++j) {
-        const int i = tid + j * NT;
-        if (i < total4) stg[j] = src4[i];
-      }
-    }
-    // ---- row norms (wave-parallel: 4 lanes per row) ----
-    {
+    float* xt = xtbuf + cur * KM_TR * d4;
+    // prefetch next tile into the other buffer (in flight all tile long)
+    const long ntile = tile + gridDim.x;
+    const bool pf = use_glds && ntile < tiles && (m - ntile * KM_TR) >= KM_TR;
+    if (pf)
+      glds_tile(x + ntile * KM_TR * (long)d, xtbuf + (cur ^ 1) * KM_TR * d4,
+                KM_TR * d4, wave, lane);
+    // ---- row norms: only the dists output (UPDATE=0) needs per-row ‖x‖²;
+    // the Lloyd path folds Σ‖x‖² into the update column pass instead.
+    if (!UPDATE && (PHASES & 2)) {
       const int r = tid >> 2;
       const int q = tid & 3;
       float s = 0.0f;
@@ -380,10 +448,10 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       s += __shfl_xor(s, 1, 4);
       s += __shfl_xor(s, 2, 4);
       if (r < rows && q == 0) xn[r] = s;
+      lds_barrier();
     }
-    __syncthreads();
     // ---- MFMA distance GEMM: this wave's 16 rows vs all center tiles ----
-    {
+    if (PHASES & 2) {
       const int ri = wave * 16 + lj;  // A-fragment row for this lane
       f32x4 acc[NTILES];  // statically indexed, branch-free unrolls
 #pragma unroll
@@ -393,27 +461,52 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       }
       const float* xrow = xt + ri * d4 + lk;
       const float* cb = ct + lj * d4 + lk;
-      // software pipeline: fragment loads for step k0+4 issue before the
-      // MFMAs of step k0 (one lgkmcnt wait per step instead of per MFMA)
-      float a_cur = xrow[0];
-      float b_cur[NTILES];
+      // Ping-pong 2-step software pipeline with NO register rotation: the
+      // rotation movs of the previous version forced lgkmcnt(0) (drain
+      // everything) each step; ping-pong buffers let hipcc emit counted
+      // lgkmcnt(N) so each step's fragment reads land a half-iteration
+      // before their MFMAs.
+      // BRANCHLESS body: guards inside the loop create phi joins that make
+      // the allocator ping-pong the accumulators VGPR<->AGPR (28 accvgpr
+      // moves per step); prefetch indices are clamped instead (an in-bounds
+      // read whose value goes unused in the tail costs nothing).
+      const int kl = d4 - 4;  // last valid step
+      float a0 = xrow[0];
+      float a1 = xrow[min(4, kl)];
+      float b0[NTILES], b1[NTILES];
 #pragma unroll
-      for (int t = 0; t < NTILES; ++t) b_cur[t] = cb[16 * t * d4];
-      for (int k0 = 0; k0 < d4; k0 += 4) {
-        float a_nxt = 0.0f;
-        float b_nxt[NTILES];
-        if (k0 + 4 < d4) {
-          a_nxt = xrow[k0 + 4];
+      for (int t = 0; t < NTILES; ++t) {
+        b0[t] = cb[16 * t * d4];
+        b1[t] = cb[16 * t * d4 + min(4, kl)];
+      }
+      int k0 = 0;
+      for (; k0 + 8 <= d4; k0 += 8) {
 #pragma unroll
-          for (int t = 0; t < NTILES; ++t) b_nxt[t] = cb[16 * t * d4 + k0 + 4];
+        for (int t = 0; t < NTILES; ++t)
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],
+                                                        0, 0, 0);
+        {
+          const int kn = min(k0 + 8, kl);
+          a0 = xrow[kn];
+#pragma unroll
+          for (int t = 0; t < NTILES; ++t) b0[t] = cb[16 * t * d4 + kn];
         }
 #pragma unroll
         for (int t = 0; t < NTILES; ++t)
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a_cur, b_cur[t],
-                                                        acc[t], 0, 0, 0);
-        a_cur = a_nxt;
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1[t], acc[t],
+                                                        0, 0, 0);
+        {
+          const int kn = min(k0 + 12, kl);
+          a1 = xrow[kn];
 #pragma unroll
-        for (int t = 0; t < NTILES; ++t) b_cur[t] = b_nxt[t];
+          for (int t = 0; t < NTILES; ++t) b1[t] = cb[16 * t * d4 + kn];
+        }
+      }
+      if (k0 < d4) {  // d4 % 8 == 4 tail: one more step (data in buf0)
+#pragma unroll
+        for (int t = 0; t < NTILES; ++t)
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],
+                                                        0, 0, 0);
       }
       // epilogue: per-lane running argmax over center tiles, then reduce
       // across the 16 lanes of the row group (rows (lane>>4)*4 + reg).
@@ -447,20 +540,19 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
         for (int g = 0; g < 4; ++g) {
           const int r = rbase + g;
           if (r < rows) {
-            const float d2 = fmaxf(xn[r] - 2.0f * bests[g], 0.0f);
             labels[r0 + r] = bjs[g];
             if (UPDATE) {
               rlab[r] = bjs[g];
-              iacc += (double)d2;
+              iacc += -2.0 * (double)bests[g];  // + Σ‖x‖² from update pass
             } else {
-              dists[r0 + r] = d2;
+              dists[r0 + r] = fmaxf(xn[r] - 2.0f * bests[g], 0.0f);
             }
           }
         }
       }
     }
-    if (UPDATE) {
-      __syncthreads();  // rlab ready
+    if (UPDATE && (PHASES & 4)) {
+      lds_barrier();  // rlab ready
       // Column-partitioned accumulation: wave w owns columns
       // [w*dq, (w+1)*dq); each column has exactly ONE owner lane, so plain
       // LDS read-modify-write suffices (the per-element atomicAdd version
@@ -469,12 +561,42 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       const int dq = (d + 3) >> 2;  // columns per wave
       const int c = wave * dq + lane;
       if (lane < dq && c < d) {
-        for (int r = 0; r < rows; ++r) {
-          bins[rlab[r] * d + c] += xt[r * d4 + c];
+        float sx2 = 0.0f;
+        int r = 0;
+        for (; r + 4 <= rows; r += 4) {
+          const int l0 = min(max(rlab[r], 0), k - 1);
+          const int l1 = min(max(rlab[r + 1], 0), k - 1);
+          const int l2 = min(max(rlab[r + 2], 0), k - 1);
+          const int l3 = min(max(rlab[r + 3], 0), k - 1);
+          const float v0 = xt[r * d4 + c];
+          const float v1 = xt[(r + 1) * d4 + c];
+          const float v2 = xt[(r + 2) * d4 + c];
+          const float v3 = xt[(r + 3) * d4 + c];
+          bins[l0 * d + c] += v0;
+          bins[l1 * d + c] += v1;
+          bins[l2 * d + c] += v2;
+          bins[l3 * d + c] += v3;
+          sx2 += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;
         }
+        for (; r < rows; ++r) {
+          const int lr = min(max(rlab[r], 0), k - 1);
+          const float v = xt[r * d4 + c];
+          bins[lr * d + c] += v;
+          sx2 += v * v;
+        }
+        iacc += (double)sx2;
       }
-      if (tid < rows) atomicAdd(&bcnt[rlab[tid]], 1.0f);
+      if (tid < rows) atomicAdd(&bcnt[min(max(rlab[tid], 0), k - 1)], 1.0f);
     }
+    // retire this tile: prefetched DMA landed (or stage the next tile
+    // directly when it is partial / glds unavailable), then swap buffers
+    if (pf) {
+      vm_drain();
+    } else if ((PHASES & 1) && ntile < tiles) {
+      stage_direct(xtbuf + (cur ^ 1) * KM_TR * d4, ntile);
+    }
+    lds_barrier();
+    cur ^= 1;
   }
 
   if (UPDATE) {
@@ -531,7 +653,7 @@ extern "C" __global__ __launch_bounds__(NT) void affine_transform_k(
 static size_t kmeans_smem_bytes_impl(int k, int d, int update) {
   const int d4 = (d + 3) & ~3;
   const int k16 = (k + 15) & ~15;
-  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)64 * d4 + 2 * 64;
+  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)2 * 64 * d4 + 2 * 64;
   if (update) f32s += (size_t)k * d + k;
   return (((f32s * 4 + 15) & ~15)) + NT * sizeof(double);
 }
@@ -581,23 +703,26 @@ void launch_glm_value_grad(const float* x, const float* y, const float* w,
   // d%4==0: contiguous LDS image (float4 staging, <=2-way bank conflicts);
   // otherwise odd stride (conflict-free, scalar staging).
   const int ld = ((d & 3) == 0) ? d : (d | 1);
-  const size_t smem =
-      ((((((d + 3) & ~3) + GLM_TR * ld + 2 * GLM_TR) * sizeof(float) + 15) &
-        ~15)) + 2 * NT * sizeof(double);
+  auto smem_for = [&](int nbuf) {
+    return ((((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
+              sizeof(float) + 15) & ~15)) + 2 * NT * sizeof(double);
+  };
+  const int nbuf = (smem_for(2) <= 158 * 1024) ? 2 : 1;
+  const size_t smem = smem_for(nbuf);
   switch (family) {
     case 0:
       hipLaunchKernelGGL((glm_value_grad_k<0>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, loss_slab, gw_slab,
+                         stream, x, y, w, b, m, d, ld, nbuf, loss_slab, gw_slab,
                          gb_slab);
       break;
     case 1:
       hipLaunchKernelGGL((glm_value_grad_k<1>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, loss_slab, gw_slab,
+                         stream, x, y, w, b, m, d, ld, nbuf, loss_slab, gw_slab,
                          gb_slab);
       break;
     default:
       hipLaunchKernelGGL((glm_value_grad_k<2>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, loss_slab, gw_slab,
+                         stream, x, y, w, b, m, d, ld, nbuf, loss_slab, gw_slab,
                          gb_slab);
   }
 }
@@ -612,6 +737,22 @@ void launch_kmeans_assign(const float* x, const float* cen, long m, int d,
                           hipStream_t stream) {
   kmeans_dispatch<1>(x, cen, m, d, k, labels, nullptr, sums_slab, cnt_slab,
                      inertia_slab, nblocks, stream);
+}
+
+void launch_kmeans_probe(const float* x, const float* cen, long m, int d,
+                         int k, int* labels, float* sums_slab,
+                         float* cnt_slab, double* inertia_slab, int nblocks,
+                         int phases, hipStream_t stream) {
+  const size_t smem = kmeans_smem_bytes_impl(k, d, 1);
+  if (((k + 15) & ~15) / 16 != 7) return;  // probe supports k=100..112 only
+#define KM_P(PH)                                                            \
+  case PH:                                                                  \
+    hipLaunchKernelGGL((kmeans_assign_k<1, 7, PH>), dim3(nblocks), dim3(NT),\
+                       smem, stream, x, cen, m, d, k, labels, nullptr,      \
+                       sums_slab, cnt_slab, inertia_slab);                  \
+    break
+  switch (phases) { KM_P(0); KM_P(1); KM_P(2); KM_P(3); KM_P(5); KM_P(7); }
+#undef KM_P
 }
 
 void launch_closest_center(const float* x, const float* cen, long m, int d,
