@@ -1,0 +1,151 @@
+"""IncrementalPCA (dask_ml/decomposition/incremental_pca.py parity).
+
+The reference threads one model object through a sequential chain of
+per-block partial_fit tasks. MI355X mapping: each rank runs
+partial_fit over its local row blocks in rank order, handing the (tiny)
+model state to the next rank with one broadcast — W broadcasts of a few
+k·d floats instead of moving any data blocks.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from ..base import BaseEstimator, TransformerMixin, check_array, check_is_fitted
+from ..core.array import DistArray
+
+
+class IncrementalPCA(BaseEstimator, TransformerMixin):
+    def __init__(self, n_components: Optional[int] = None,
+                 whiten: bool = False, batch_size: Optional[int] = None):
+        self.n_components = n_components
+        self.whiten = whiten
+        self.batch_size = batch_size
+
+    def _reset(self, d: int, device):
+        self.components_ = None
+        self.singular_values_ = None
+        self.mean_ = torch.zeros(d, dtype=torch.float64, device=device)
+        self.var_ = torch.zeros(d, dtype=torch.float64, device=device)
+        self.n_samples_seen_ = 0
+
+    def _partial_fit_block(self, xb: torch.Tensor, k: int):
+        """sklearn IncrementalPCA.partial_fit math on one block (float64)."""
+        x = xb.to(torch.float64)
+        nb = x.shape[0]
+        if nb == 0:
+            return
+        n_old = self.n_samples_seen_
+        n_tot = n_old + nb
+        batch_mean = x.mean(dim=0)
+        batch_var = x.var(dim=0, unbiased=False)
+        # incremental mean/var (Chan)
+        new_mean = (self.mean_ * n_old + batch_mean * nb) / n_tot
+        delta = self.mean_ - batch_mean
+        new_var = (self.var_ * n_old + batch_var * nb
+                   + (n_old * nb / n_tot) * delta * delta) / n_tot
+        if n_old == 0:
+            stack = x - batch_mean
+        else:
+            corr = math.sqrt(n_old * nb / n_tot) * (self.mean_ - batch_mean)
+            stack = torch.cat([
+                self.singular_values_[:, None] * self.components_,
+                x - batch_mean,
+                corr[None, :],
+            ], dim=0)
+        _, s, vh = torch.linalg.svd(stack, full_matrices=False)
+        idx = vh.abs().argmax(dim=1)
+        signs = torch.sign(vh[torch.arange(vh.shape[0], device=vh.device),
+                               idx])
+        signs[signs == 0] = 1.0
+        vh = vh * signs[:, None]
+        self.components_ = vh[:k]
+        self.singular_values_ = s[:k]
+        self.mean_ = new_mean
+        self.var_ = new_var
+        self.n_samples_seen_ = n_tot
+
+    def _bcast_state(self, comm, src: int, d: int, k: int):
+        if comm.world_size == 1:
+            return
+        dev = comm.device
+        buf = torch.zeros(k * d + k + 2 * d + 1, dtype=torch.float64,
+                          device=dev)
+        if comm.rank == src and self.components_ is not None:
+            buf[:k * d] = self.components_.reshape(-1)
+            buf[k * d:k * d + k] = self.singular_values_
+            buf[k * d + k:k * d + k + d] = self.mean_
+            buf[k * d + k + d:k * d + k + 2 * d] = self.var_
+            buf[-1] = float(self.n_samples_seen_)
+        comm.broadcast(buf, src=src)
+        if comm.rank != src:
+            self.components_ = buf[:k * d].reshape(k, d).clone()
+            self.singular_values_ = buf[k * d:k * d + k].clone()
+            self.mean_ = buf[k * d + k:k * d + k + d].clone()
+            self.var_ = buf[k * d + k + d:k * d + k + 2 * d].clone()
+            self.n_samples_seen_ = int(buf[-1])
+
+    def fit(self, X, y=None):
+        X = check_array(X)
+        d = X.ncols
+        k = self.n_components or min(X.global_rows, d)
+        bs = self.batch_size or max(5 * d, 100)
+        self._reset(d, X.device)
+        comm = X.comm
+        for r in range(comm.world_size):
+            if comm.rank == r:
+                for blk in X.blocks(bs):
+                    self._partial_fit_block(blk, k)
+            self._bcast_state(comm, r, d, k)
+        self.n_components_ = k
+        self.n_features_in_ = d
+        n = self.n_samples_seen_
+        self.explained_variance_ = (
+            self.singular_values_ ** 2 / max(n - 1, 1)).to(X.dtype)
+        total_var = float(self.var_.sum()) * n / max(n - 1, 1)
+        self.explained_variance_ratio_ = (
+            self.explained_variance_.double() / max(total_var, 1e-300)
+        ).to(X.dtype)
+        self.components_ = self.components_.to(X.dtype)
+        self.singular_values_ = self.singular_values_.to(X.dtype)
+        return self
+
+    def partial_fit(self, X, y=None):
+        X = check_array(X)
+        d = X.ncols
+        k = self.n_components or d
+        if not hasattr(self, "n_samples_seen_"):
+            self._reset(d, X.device)
+        # each rank folds its shard in rank order (sequential chain parity)
+        comm = X.comm
+        if self.components_ is not None and self.components_.dtype != torch.float64:
+            self.components_ = self.components_.double()
+            self.singular_values_ = self.singular_values_.double()
+        for r in range(comm.world_size):
+            if comm.rank == r:
+                self._partial_fit_block(X.local if X.ndim == 2
+                                        else X.local[:, None], k)
+            self._bcast_state(comm, r, d, k)
+        self.n_components_ = k
+        self.n_features_in_ = d
+        return self
+
+    def transform(self, X):
+        check_is_fitted(self, "components_")
+        X = check_array(X)
+        V = self.components_.to(device=X.device, dtype=X.dtype)
+        mean = self.mean_.to(device=X.device, dtype=X.dtype)
+        return X.map_local(lambda t: (t - mean) @ V.t())
+
+    def fit_transform(self, X, y=None):
+        return self.fit(X).transform(X)
+
+    def inverse_transform(self, X):
+        check_is_fitted(self, "components_")
+        X = check_array(X, dtype=None)
+        V = self.components_.to(device=X.device)
+        mean = self.mean_.to(device=X.device)
+        return X.map_local(lambda t: t @ V.to(t.dtype) + mean.to(t.dtype))

@@ -1,0 +1,122 @@
+"""PCA (dask_ml/decomposition/pca.py parity — SURVEY §2.1).
+
+Mean-center (fused column-stats pass + one small allreduce) then SVD:
+svd_solver 'full'/'tsqr' -> distributed TSQR (ops/linalg.tsqr);
+'randomized' -> Halko range finder (ops/linalg.svd_compressed);
+'auto' -> randomized when n_components < 0.8·ncols, else tsqr.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ..base import BaseEstimator, TransformerMixin, check_array, check_is_fitted
+from ..core.array import DistArray
+from ..ops import affine_transform
+from ..ops.linalg import svd_compressed, tsqr
+
+
+class PCA(BaseEstimator, TransformerMixin):
+    def __init__(self, n_components: Optional[int] = None,
+                 whiten: bool = False, svd_solver: str = "auto",
+                 iterated_power: int = 2, random_state: int = 0):
+        self.n_components = n_components
+        self.whiten = whiten
+        self.svd_solver = svd_solver
+        self.iterated_power = iterated_power
+        self.random_state = random_state
+
+    def _solver(self, d: int, k: int) -> str:
+        if self.svd_solver != "auto":
+            return self.svd_solver
+        return "randomized" if k < 0.8 * d else "full"
+
+    def _fit(self, X: DistArray):
+        d = X.ncols
+        n = X.global_rows
+        k = self.n_components if self.n_components is not None else min(n, d)
+        mean = X.mean(axis=0).to(X.dtype)
+        ones = torch.ones(d, dtype=X.dtype, device=X.device)
+        Xc = X.map_local(lambda t: affine_transform(t, ones, -mean))
+        solver = self._solver(d, k)
+        if solver in ("full", "tsqr"):
+            U, s, Vh = tsqr(Xc, compute_svd=True)
+            Uk = DistArray(U._as2d()[:, :k], X.global_rows, X.row_offset,
+                           X.comm)
+            s_all = s
+            s = s[:k]
+            Vh = Vh[:k]
+        elif solver == "randomized":
+            U, s, Vh = svd_compressed(
+                Xc, k, n_power_iter=self.iterated_power,
+                seed=int(self.random_state or 0))
+            Uk = U
+            s_all = None
+        else:
+            raise ValueError(f"unknown svd_solver {solver!r}")
+        n_f = float(n)
+        exp_var = (s.double() ** 2) / max(n_f - 1, 1.0)
+        # total variance from the exact column moments (all components)
+        _, _, m2 = Xc.moments()
+        total_var = float(m2.sum() / max(n_f - 1, 1.0))
+        self.mean_ = mean
+        self.components_ = Vh
+        self.singular_values_ = s
+        self.explained_variance_ = exp_var.to(X.dtype)
+        self.explained_variance_ratio_ = (
+            (exp_var / max(total_var, 1e-300)).to(X.dtype))
+        if k < min(n, d):
+            self.noise_variance_ = (
+                (total_var - float(exp_var.sum())) / (min(n, d) - k))
+        else:
+            self.noise_variance_ = 0.0
+        self.n_components_ = k
+        self.n_features_in_ = d
+        self.n_samples_ = n
+        return Uk, s, Vh
+
+    def fit(self, X, y=None):
+        X = check_array(X)
+        self._fit(X)
+        return self
+
+    def fit_transform(self, X, y=None):
+        X = check_array(X)
+        U, s, _ = self._fit(X)
+        if self.whiten:
+            import math
+            scale = math.sqrt(max(X.global_rows - 1, 1))
+            return U.map_local(lambda t: t * scale)
+        return U.map_local(lambda t: t * s.to(t.dtype))
+
+    def transform(self, X):
+        check_is_fitted(self, "components_")
+        X = check_array(X)
+        V = self.components_.to(X.device)
+        mean = self.mean_.to(X.device)
+        if self.whiten:
+            inv = (1.0 / self.explained_variance_.to(X.device).sqrt())
+
+            def _t(t):
+                return ((t - mean) @ V.to(t.dtype).t()) * inv.to(t.dtype)
+        else:
+            def _t(t):
+                return (t - mean) @ V.to(t.dtype).t()
+        return X.map_local(_t)
+
+    def inverse_transform(self, X):
+        check_is_fitted(self, "components_")
+        X = check_array(X, dtype=None)
+        V = self.components_.to(X.device)
+        mean = self.mean_.to(X.device)
+        if self.whiten:
+            sc = self.explained_variance_.to(X.device).sqrt()
+
+            def _t(t):
+                return (t * sc.to(t.dtype)) @ V.to(t.dtype) + mean
+        else:
+            def _t(t):
+                return t @ V.to(t.dtype) + mean
+        return X.map_local(_t)
