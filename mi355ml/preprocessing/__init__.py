@@ -1,3 +1,11 @@
-from .data import StandardScaler, MinMaxScaler
+from .data import (MinMaxScaler, Normalizer, PolynomialFeatures,
+                   QuantileTransformer, RobustScaler, StandardScaler)
+from .label import LabelEncoder
+from ._encoders import OneHotEncoder, OrdinalEncoder
+from ._block_transformer import BlockTransformer
 
-__all__ = ["StandardScaler", "MinMaxScaler"]
+__all__ = [
+    "StandardScaler", "MinMaxScaler", "RobustScaler", "QuantileTransformer",
+    "Normalizer", "PolynomialFeatures", "LabelEncoder", "OneHotEncoder",
+    "OrdinalEncoder", "BlockTransformer",
+]
