@@ -1,0 +1,237 @@
+"""Adaptive incremental hyperparameter search
+(dask_ml/model_selection/_incremental.py parity — SURVEY §3.5).
+
+The reference keeps one sklearn model per candidate alive on the workers as
+futures and drives rounds of {partial_fit on one scattered block, score on a
+validation block, _adapt}. MI355X mapping: trials are pinned
+round-robin to ranks; each rank partial_fits its trials on slabs of its OWN
+shard (data is already where the trials are — no scatter), scores on its
+local validation shard, and rounds synchronize with one tiny
+all_gather_object of (trial_id, score) pairs; the _adapt decision runs
+redundantly+deterministically on every rank.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+from sklearn.base import clone
+from sklearn.model_selection import ParameterSampler
+
+from .. import config
+from ..base import BaseEstimator, check_array
+from ..core.array import DistArray
+from ..core import random as mlrand
+from ._split import train_test_split
+
+
+class _Trial:
+    __slots__ = ("tid", "params", "model", "history", "pf_calls",
+                 "best_score", "since_best", "score")
+
+    def __init__(self, tid, params):
+        self.tid = tid
+        self.params = params
+        self.model = None
+        self.history: List[float] = []
+        self.pf_calls = 0
+        self.best_score = -math.inf
+        self.since_best = 0
+        self.score = -math.inf
+
+
+class BaseIncrementalSearchCV(BaseEstimator):
+    def __init__(self, estimator, parameters, n_initial_parameters=10,
+                 test_size=0.15, patience=False, tol=1e-3, max_iter=100,
+                 random_state=0, scores_per_fit=1, verbose=False):
+        self.estimator = estimator
+        self.parameters = parameters
+        self.n_initial_parameters = n_initial_parameters
+        self.test_size = test_size
+        self.patience = patience
+        self.tol = tol
+        self.max_iter = max_iter
+        self.random_state = random_state
+        self.scores_per_fit = scores_per_fit
+        self.verbose = verbose
+
+    # -- subclass hook ----------------------------------------------------
+    def _adapt(self, trials: List[_Trial], round_idx: int) -> Dict[int, int]:
+        """Return {tid: n_partial_fits_next_round} for survivors."""
+        raise NotImplementedError
+
+    def _initial_candidates(self) -> List[Dict]:
+        n = self.n_initial_parameters
+        if n == "grid":
+            from sklearn.model_selection import ParameterGrid
+            return list(ParameterGrid(self.parameters))
+        return list(ParameterSampler(self.parameters, n,
+                                     random_state=self.random_state))
+
+    # -- machinery --------------------------------------------------------
+    def _slabs(self, X: DistArray, y: DistArray):
+        bs = int(config.get("array.block_rows"))
+        xs, ys = [], []
+        x2 = X.local
+        m = x2.shape[0]
+        for lo in range(0, max(m, 1), bs):
+            xe = x2[lo:lo + bs]
+            ye = y.local[lo:lo + bs]
+            if xe.shape[0]:
+                xs.append(xe.cpu().numpy())
+                ys.append(ye.cpu().numpy())
+        if not xs:
+            xs, ys = [x2.cpu().numpy()], [y.local.cpu().numpy()]
+        return xs, ys
+
+    def fit(self, X, y=None, classes=None, **fit_params):
+        X = check_array(X, dtype=None)
+        y = check_array(y, dtype=None)
+        comm = X.comm
+        Xtr, Xte, ytr, yte = train_test_split(
+            X, y, test_size=self.test_size,
+            random_state=int(self.random_state or 0))
+        xs, ys = self._slabs(Xtr, ytr)
+        xval = Xte.local.cpu().numpy()
+        yval = yte.local.cpu().numpy()
+        if classes is None and hasattr(clone(self.estimator), "partial_fit"):
+            if getattr(self.estimator, "_estimator_type", "") == "classifier":
+                from ..ops.unique import unique
+                classes = unique(y).cpu().numpy()
+
+        cands = self._initial_candidates()
+        trials = [_Trial(i, p) for i, p in enumerate(cands)]
+        owned = {t.tid: t for t in trials if t.tid % comm.world_size == comm.rank}
+        for t in owned.values():
+            t.model = clone(self.estimator).set_params(**t.params)
+
+        budgets = {t.tid: 1 for t in trials}
+        round_idx = 0
+        meta_pf_calls = 0
+        while budgets:
+            # run this round's partial_fits + scores for my trials
+            local_scores = []
+            for tid, n_pf in budgets.items():
+                t = owned.get(tid)
+                if t is None:
+                    continue
+                for _ in range(n_pf):
+                    blk = t.pf_calls % len(xs)
+                    if classes is not None and t.pf_calls == 0:
+                        t.model.partial_fit(xs[blk], ys[blk],
+                                            classes=classes, **fit_params)
+                    else:
+                        t.model.partial_fit(xs[blk], ys[blk], **fit_params)
+                    t.pf_calls += 1
+                sc = float(t.model.score(xval, yval)) if len(yval) else 0.0
+                local_scores.append((tid, sc, t.pf_calls))
+            # exchange scores; all ranks see the global picture
+            gathered = comm.all_gather_object(local_scores)
+            all_scores = {tid: (sc, pf) for part in gathered
+                          for tid, sc, pf in part}
+            for t in trials:
+                if t.tid in all_scores:
+                    sc, pf = all_scores[t.tid]
+                    t.score = sc
+                    t.pf_calls = pf
+                    t.history.append(sc)
+                    meta_pf_calls += pf
+                    if sc > t.best_score + self.tol:
+                        t.best_score = sc
+                        t.since_best = 0
+                    else:
+                        t.since_best += 1
+            round_idx += 1
+            budgets = self._adapt(
+                [t for t in trials if t.tid in all_scores], round_idx)
+            # drop dead trials' models
+            owned = {tid: t for tid, t in owned.items() if tid in budgets}
+
+        # best trial
+        alive = [t for t in trials if t.history]
+        best = max(alive, key=lambda t: (t.score, -t.tid))
+        self.best_index_ = best.tid
+        self.best_params_ = best.params
+        self.best_score_ = best.score
+        owner = best.tid % comm.world_size
+        model = None
+        if comm.rank == owner:
+            model = owned.get(best.tid)
+            model = model.model if model is not None else None
+        if model is None and comm.rank == owner:
+            # owner dropped it after the final round: refit quickly
+            model = clone(self.estimator).set_params(**best.params)
+            for xb, yb in zip(xs, ys):
+                if classes is not None:
+                    model.partial_fit(xb, yb, classes=classes)
+                else:
+                    model.partial_fit(xb, yb)
+        self.best_estimator_ = comm.broadcast_object(model, src=owner)
+        self.cv_results_ = {
+            "params": [t.params for t in trials],
+            "test_score": np.array([t.score for t in trials]),
+            "partial_fit_calls": np.array([t.pf_calls for t in trials]),
+            "model_id": np.array([t.tid for t in trials]),
+        }
+        self.history_ = [
+            {"model_id": t.tid, "params": t.params, "scores": t.history}
+            for t in trials]
+        self.n_iter_ = round_idx
+        self.metadata_ = {"n_models": len(trials),
+                          "partial_fit_calls": meta_pf_calls}
+        return self
+
+    def predict(self, X):
+        from ..wrappers import _blockwise_method
+        return _blockwise_method(self.best_estimator_, "predict",
+                                 check_array(X, dtype=None))
+
+    def predict_proba(self, X):
+        from ..wrappers import _blockwise_method
+        return _blockwise_method(self.best_estimator_, "predict_proba",
+                                 check_array(X, dtype=None))
+
+    def score(self, X, y):
+        from ..metrics import accuracy_score, r2_score
+        yp = self.predict(X)
+        y = check_array(y, dtype=None)
+        if getattr(self.best_estimator_, "_estimator_type", "") == "regressor":
+            return r2_score(y, yp)
+        return accuracy_score(y, yp)
+
+
+class IncrementalSearchCV(BaseIncrementalSearchCV):
+    """Adaptive trainer with 'inverse' decay of the candidate population
+    (keep top ~n0/(t+1)^decay_rate each round)."""
+
+    def __init__(self, estimator, parameters, n_initial_parameters=10,
+                 decay_rate=1.0, test_size=0.15, patience=False, tol=1e-3,
+                 max_iter=100, random_state=0, fits_per_score=1,
+                 verbose=False):
+        super().__init__(estimator, parameters, n_initial_parameters,
+                         test_size, patience, tol, max_iter, random_state,
+                         verbose=verbose)
+        self.decay_rate = decay_rate
+        self.fits_per_score = fits_per_score
+
+    def _adapt(self, trials, round_idx):
+        alive = sorted(trials, key=lambda t: -t.score)
+        if self.decay_rate:
+            keep = max(1, int(math.ceil(
+                self.n_initial_parameters
+                / (round_idx + 1) ** self.decay_rate)))
+            alive = alive[:keep]
+        if self.patience:
+            alive = [t for t in alive if t.since_best < self.patience] or \
+                alive[:1]
+        out = {}
+        for t in alive:
+            if t.pf_calls < self.max_iter:
+                out[t.tid] = self.fits_per_score
+        if len(alive) <= 1 and all(t.pf_calls >= self.max_iter
+                                   for t in alive):
+            return {}
+        return out

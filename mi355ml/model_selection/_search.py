@@ -1,0 +1,185 @@
+"""GridSearchCV / RandomizedSearchCV (dask_ml/model_selection/_search.py
+parity).
+
+The reference's distinguishing feature is building one task graph for the
+whole CV x candidate grid with shared pipeline prefixes deduplicated by
+token. The eager SPMD equivalent: candidates run as
+distributed fits (every rank participates in each fit over the row-sharded
+folds), and for Pipeline estimators the fit_transform of each distinct
+(prefix-params, fold) pair is computed ONCE and cached by token — a vec
+shared by all clf-grid candidates is fitted once per fold, not once per
+candidate.
+"""
+
+from __future__ import annotations
+
+import numbers
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+from sklearn.base import clone
+from sklearn.model_selection import ParameterGrid, ParameterSampler
+
+from ..base import BaseEstimator, check_array
+from ..core.array import DistArray
+from ._split import check_cv, mask_take
+
+
+def _tokenize(obj: Any) -> str:
+    """Deterministic content token (the reference's dask.base.tokenize
+    analog, dask/base.py:789) for params/estimators."""
+    import hashlib
+    import json
+
+    def norm(o):
+        if isinstance(o, dict):
+            return {str(k): norm(v) for k, v in sorted(o.items())}
+        if isinstance(o, (list, tuple)):
+            return [norm(v) for v in o]
+        if isinstance(o, (str, bool, numbers.Number)) or o is None:
+            return o
+        if hasattr(o, "get_params"):
+            return {"__est__": type(o).__name__, **norm(o.get_params(deep=False))}
+        return repr(o)
+
+    return hashlib.sha1(
+        json.dumps(norm(obj), sort_keys=True, default=repr).encode()
+    ).hexdigest()
+
+
+def _is_pipeline(est) -> bool:
+    return hasattr(est, "steps") and isinstance(getattr(est, "steps"), list)
+
+
+class _SearchBase(BaseEstimator):
+    def __init__(self, estimator, scoring=None, cv=None, refit=True,
+                 return_train_score=False):
+        self.estimator = estimator
+        self.scoring = scoring
+        self.cv = cv
+        self.refit = refit
+        self.return_train_score = return_train_score
+
+    def _candidates(self) -> List[Dict]:
+        raise NotImplementedError
+
+    def _score(self, est, Xte, yte) -> float:
+        if self.scoring is not None:
+            from ..metrics.scorer import check_scoring
+            scorer = check_scoring(est, self.scoring)
+            return scorer(est, Xte, yte)
+        return est.score(Xte, yte) if yte is not None else est.score(Xte)
+
+    def _fit_candidate(self, params, Xtr, ytr, Xte, yte, cache, fold):
+        est = clone(self.estimator)
+        est.set_params(**params)
+        if _is_pipeline(est) and len(est.steps) > 1:
+            # pipeline-prefix dedup: fit_transform shared prefixes once
+            Xtr_c, Xte_c = Xtr, Xte
+            prefix_token = str(fold)
+            steps = est.steps
+            fitted_steps = []
+            for name, step in steps[:-1]:
+                prefix_token = _tokenize([prefix_token, name, step])
+                hit = cache.get(prefix_token)
+                if hit is None:
+                    stf = clone(step).fit(Xtr_c, ytr)
+                    Xtr_n = stf.transform(Xtr_c)
+                    Xte_n = stf.transform(Xte_c)
+                    cache[prefix_token] = (stf, Xtr_n, Xte_n)
+                stf, Xtr_c, Xte_c = cache[prefix_token]
+                fitted_steps.append((name, stf))
+            final_name, final = steps[-1]
+            finalf = clone(final)
+            if ytr is not None:
+                finalf.fit(Xtr_c, ytr)
+            else:
+                finalf.fit(Xtr_c)
+            score = self._score(finalf, Xte_c, yte)
+            return score
+        if ytr is not None:
+            est.fit(Xtr, ytr)
+        else:
+            est.fit(Xtr)
+        return self._score(est, Xte, yte)
+
+    def fit(self, X, y=None, **fit_params):
+        X = check_array(X, dtype=None)
+        if y is not None:
+            y = check_array(y, dtype=None)
+        cv = check_cv(self.cv)
+        candidates = self._candidates()
+        n_splits = cv.get_n_splits(X, y)
+        scores = np.zeros((len(candidates), n_splits))
+        for f, (tr_mask, te_mask) in enumerate(cv.split(X, y)):
+            arrays = [X] + ([y] if y is not None else [])
+            tr = mask_take(arrays, tr_mask)
+            te = mask_take(arrays, te_mask)
+            Xtr, Xte = tr[0], te[0]
+            ytr = tr[1] if y is not None else None
+            yte = te[1] if y is not None else None
+            cache: Dict[str, Any] = {}
+            for c, params in enumerate(candidates):
+                scores[c, f] = self._fit_candidate(
+                    params, Xtr, ytr, Xte, yte, cache, f)
+        mean = scores.mean(axis=1)
+        std = scores.std(axis=1)
+        best = int(np.argmax(mean))
+        self.cv_results_ = {
+            "params": candidates,
+            "mean_test_score": mean,
+            "std_test_score": std,
+            "rank_test_score": ((-mean).argsort().argsort() + 1).astype(int),
+        }
+        for f in range(n_splits):
+            self.cv_results_[f"split{f}_test_score"] = scores[:, f]
+        self.best_index_ = best
+        self.best_params_ = candidates[best]
+        self.best_score_ = float(mean[best])
+        if self.refit:
+            est = clone(self.estimator)
+            est.set_params(**self.best_params_)
+            if y is not None:
+                est.fit(X, y)
+            else:
+                est.fit(X)
+            self.best_estimator_ = est
+        return self
+
+    def predict(self, X):
+        return self.best_estimator_.predict(X)
+
+    def predict_proba(self, X):
+        return self.best_estimator_.predict_proba(X)
+
+    def transform(self, X):
+        return self.best_estimator_.transform(X)
+
+    def score(self, X, y=None):
+        return self._score(self.best_estimator_, check_array(X, dtype=None),
+                           check_array(y, dtype=None) if y is not None
+                           else None)
+
+
+class GridSearchCV(_SearchBase):
+    def __init__(self, estimator, param_grid, scoring=None, cv=None,
+                 refit=True, return_train_score=False):
+        super().__init__(estimator, scoring, cv, refit, return_train_score)
+        self.param_grid = param_grid
+
+    def _candidates(self):
+        return list(ParameterGrid(self.param_grid))
+
+
+class RandomizedSearchCV(_SearchBase):
+    def __init__(self, estimator, param_distributions, n_iter=10,
+                 scoring=None, cv=None, refit=True, random_state=0,
+                 return_train_score=False):
+        super().__init__(estimator, scoring, cv, refit, return_train_score)
+        self.param_distributions = param_distributions
+        self.n_iter = n_iter
+        self.random_state = random_state
+
+    def _candidates(self):
+        return list(ParameterSampler(self.param_distributions, self.n_iter,
+                                     random_state=self.random_state))

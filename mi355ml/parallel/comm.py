@@ -100,6 +100,21 @@ class Communicator:
         if self.world_size > 1:
             dist.barrier(group=self.group)
 
+    # -- control-plane object collectives (tiny, pickle-based) ------------
+    def broadcast_object(self, obj, src: int = 0):
+        if self.world_size == 1:
+            return obj
+        lst = [obj if self.rank == src else None]
+        dist.broadcast_object_list(lst, src=src, group=self.group)
+        return lst[0]
+
+    def all_gather_object(self, obj) -> list:
+        if self.world_size == 1:
+            return [obj]
+        out = [None] * self.world_size
+        dist.all_gather_object(out, obj, group=self.group)
+        return out
+
     # -- helpers ----------------------------------------------------------
     def shard_bounds(self, n_rows: int) -> tuple:
         """Contiguous row range [lo, hi) owned by this rank (balanced)."""

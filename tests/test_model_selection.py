@@ -1,0 +1,162 @@
+"""model_selection: splits, searches, wrappers, Hyperband."""
+
+import numpy as np
+import pytest
+from sklearn.linear_model import SGDClassifier
+from sklearn.pipeline import Pipeline
+
+from mi355ml.core.array import from_numpy
+from mi355ml.linear_model import LogisticRegression
+from mi355ml.model_selection import (GridSearchCV, HyperbandSearchCV,
+                                     IncrementalSearchCV, KFold,
+                                     RandomizedSearchCV,
+                                     SuccessiveHalvingSearchCV,
+                                     train_test_split)
+from mi355ml.preprocessing import StandardScaler
+from mi355ml.wrappers import Incremental, ParallelPostFit
+
+
+def _cls(n=3000, f=6, seed=0):
+    rng = np.random.default_rng(seed)
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    w = rng.standard_normal(f) * 2
+    p = 1 / (1 + np.exp(-(X @ w)))
+    y = (rng.uniform(size=n) < p).astype(np.float32)
+    return X, y
+
+
+def test_train_test_split():
+    X, y = _cls()
+    Xtr, Xte, ytr, yte = train_test_split(from_numpy(X), from_numpy(y),
+                                          test_size=0.3, random_state=0)
+    assert Xtr.global_rows + Xte.global_rows == X.shape[0]
+    assert ytr.global_rows == Xtr.global_rows
+    frac = Xte.global_rows / X.shape[0]
+    assert 0.25 < frac < 0.35
+
+
+def test_kfold():
+    X, y = _cls(n=1000)
+    Xd = from_numpy(X)
+    kf = KFold(n_splits=4)
+    sizes = []
+    for tr, te in kf.split(Xd):
+        assert bool((tr ^ te).all())
+        sizes.append(int(te.sum()))
+    assert sum(sizes) == 1000
+
+
+def test_grid_search_logreg():
+    X, y = _cls()
+    gs = GridSearchCV(LogisticRegression(solver="lbfgs", max_iter=50),
+                      {"C": [0.01, 1.0, 100.0]}, cv=3)
+    gs.fit(from_numpy(X), from_numpy(y))
+    assert gs.best_params_["C"] in (0.01, 1.0, 100.0)
+    assert gs.best_score_ > 0.7
+    assert len(gs.cv_results_["mean_test_score"]) == 3
+    pred = gs.predict(from_numpy(X)).gather().numpy()
+    assert pred.shape == (X.shape[0],)
+
+
+def test_grid_search_pipeline_prefix_dedup():
+    X, y = _cls(n=1500)
+    calls = {"n": 0}
+
+    class CountingScaler(StandardScaler):
+        def fit(self, X, y=None):
+            calls["n"] += 1
+            return super().fit(X, y)
+
+    pipe = Pipeline([("sc", CountingScaler()),
+                     ("clf", LogisticRegression(max_iter=30))])
+    gs = GridSearchCV(pipe, {"clf__C": [0.1, 1.0, 10.0]}, cv=2)
+    gs.fit(from_numpy(X), from_numpy(y))
+    # the scaler prefix is shared by all 3 candidates: fitted once per fold
+    # (2) + once by the refit on the full data — NOT once per candidate (7)
+    assert calls["n"] == 3, calls
+    assert gs.best_score_ > 0.6
+
+
+def test_randomized_search():
+    X, y = _cls(n=1200, seed=2)
+    rs = RandomizedSearchCV(LogisticRegression(max_iter=30),
+                            {"C": [0.01, 0.1, 1.0, 10.0, 100.0]},
+                            n_iter=3, cv=2, random_state=0)
+    rs.fit(from_numpy(X), from_numpy(y))
+    assert len(rs.cv_results_["params"]) == 3
+
+
+def test_parallel_post_fit():
+    X, y = _cls(n=1000, seed=3)
+    ppf = ParallelPostFit(SGDClassifier(loss="log_loss", max_iter=20,
+                                        random_state=0))
+    ppf.fit(from_numpy(X), from_numpy(y))
+    pred = ppf.predict(from_numpy(X)).gather().numpy()
+    assert pred.shape == (1000,)
+    acc = ppf.score(from_numpy(X), from_numpy(y))
+    assert acc > 0.6
+    proba = ppf.predict_proba(from_numpy(X)).gather().numpy()
+    assert proba.shape == (1000, 2)
+
+
+def test_incremental_wrapper():
+    X, y = _cls(n=2000, seed=4)
+    inc = Incremental(SGDClassifier(loss="log_loss", random_state=0),
+                      shuffle_blocks=False)
+    inc.fit(from_numpy(X), from_numpy(y), epochs=3)
+    acc = inc.score(from_numpy(X), from_numpy(y))
+    assert acc > 0.7
+
+
+def test_incremental_search():
+    X, y = _cls(n=2000, seed=5)
+    search = IncrementalSearchCV(
+        SGDClassifier(loss="log_loss", random_state=0),
+        {"alpha": [1e-5, 1e-4, 1e-3, 1e-2, 1e-1]},
+        n_initial_parameters=5, max_iter=10, random_state=0)
+    search.fit(from_numpy(X), from_numpy(y))
+    assert search.best_score_ > 0.6
+    assert search.metadata_["n_models"] == 5
+    pred = search.predict(from_numpy(X)).gather().numpy()
+    assert pred.shape == (2000,)
+
+
+def test_successive_halving():
+    X, y = _cls(n=2000, seed=6)
+    sha = SuccessiveHalvingSearchCV(
+        SGDClassifier(loss="log_loss", random_state=0),
+        {"alpha": [1e-5, 1e-4, 1e-3, 1e-2]},
+        n_initial_parameters=4, n_initial_iter=2, max_iter=12,
+        random_state=0)
+    sha.fit(from_numpy(X), from_numpy(y))
+    assert sha.best_score_ > 0.6
+
+
+def test_hyperband():
+    X, y = _cls(n=2000, seed=7)
+    hb = HyperbandSearchCV(
+        SGDClassifier(loss="log_loss", random_state=0),
+        {"alpha": [1e-5, 1e-4, 1e-3, 1e-2, 1e-1]},
+        max_iter=9, random_state=0)
+    meta = hb.metadata
+    assert meta["n_models"] >= 5
+    hb.fit(from_numpy(X), from_numpy(y))
+    assert hb.best_score_ > 0.6
+    assert hb.metadata_["partial_fit_calls"] > 0
+
+
+def _hyperband_world2(rank, world, X_bytes, y_bytes, n, f):
+    X = np.frombuffer(X_bytes, dtype=np.float32).reshape(n, f).copy()
+    y = np.frombuffer(y_bytes, dtype=np.float32).copy()
+    hb = HyperbandSearchCV(
+        SGDClassifier(loss="log_loss", random_state=0),
+        {"alpha": [1e-5, 1e-4, 1e-3, 1e-2]}, max_iter=4, random_state=0)
+    hb.fit(from_numpy(X), from_numpy(y))
+    assert hb.best_score_ > 0.55
+    assert hb.best_params_ is not None
+
+
+def test_hyperband_world2(mp_runner):
+    X, y = _cls(n=1600, seed=8)
+    mp_runner(_hyperband_world2, world=2,
+              args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1]))
