@@ -13,13 +13,17 @@ __version__ = "0.1.0"
 
 from . import config
 from .core.array import DistArray, from_numpy, shard_rows
+from .core.bag import DistList
 from .parallel import comm
 
+from . import (cluster, datasets, decomposition, diagnostics, ensemble,
+               feature_extraction, impute, linear_model, metrics,
+               model_selection, naive_bayes, preprocessing, utils, wrappers)
+
 __all__ = [
-    "config",
-    "comm",
-    "DistArray",
-    "from_numpy",
-    "shard_rows",
+    "config", "comm", "DistArray", "DistList", "from_numpy", "shard_rows",
+    "cluster", "datasets", "decomposition", "diagnostics", "ensemble",
+    "feature_extraction", "impute", "linear_model", "metrics",
+    "model_selection", "naive_bayes", "preprocessing", "utils", "wrappers",
     "__version__",
 ]

@@ -1,3 +1,4 @@
 from .k_means import KMeans
+from .spectral import SpectralClustering
 
-__all__ = ["KMeans"]
+__all__ = ["KMeans", "SpectralClustering"]

@@ -1,0 +1,109 @@
+"""Ensemble, text feature extraction, spectral clustering, utils,
+diagnostics."""
+
+import numpy as np
+import scipy.sparse as sp
+import torch
+from sklearn.linear_model import SGDClassifier, SGDRegressor
+
+import mi355ml
+from mi355ml.core.array import from_numpy
+from mi355ml.core.bag import DistList
+from mi355ml.cluster import SpectralClustering
+from mi355ml.ensemble import (BlockwiseVotingClassifier,
+                              BlockwiseVotingRegressor)
+from mi355ml.feature_extraction.text import (CountVectorizer, FeatureHasher,
+                                             HashingVectorizer)
+
+
+def test_blockwise_voting_classifier():
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((2000, 5)).astype(np.float32)
+    w = rng.standard_normal(5)
+    y = (X @ w > 0).astype(np.float32)
+    with mi355ml.config.set({"array.block_rows": 500}):
+        clf = BlockwiseVotingClassifier(
+            SGDClassifier(loss="log_loss", random_state=0, max_iter=20))
+        clf.fit(from_numpy(X), from_numpy(y))
+        assert len(clf.estimators_) == 4
+        acc = clf.score(from_numpy(X), from_numpy(y))
+        assert acc > 0.8
+        proba = clf.predict_proba(from_numpy(X)).gather().numpy()
+        assert proba.shape == (2000, 2)
+
+
+def test_blockwise_voting_regressor():
+    rng = np.random.default_rng(1)
+    X = rng.standard_normal((2000, 4)).astype(np.float32)
+    w = rng.standard_normal(4)
+    y = (X @ w).astype(np.float32)
+    with mi355ml.config.set({"array.block_rows": 1000}):
+        reg = BlockwiseVotingRegressor(SGDRegressor(random_state=0))
+        reg.fit(from_numpy(X), from_numpy(y))
+        assert reg.score(from_numpy(X), from_numpy(y)) > 0.9
+
+
+DOCS = ["the quick brown fox", "jumps over the lazy dog",
+        "the dog barks", "quick quick fox"]
+
+
+def test_hashing_vectorizer():
+    hv = HashingVectorizer(n_features=64)
+    out = hv.fit_transform(DOCS)
+    mats = out.gather()
+    m = sp.vstack(mats)
+    assert m.shape == (4, 64)
+
+
+def test_count_vectorizer():
+    cv = CountVectorizer()
+    out = cv.fit_transform(DOCS)
+    m = sp.vstack(out.gather())
+    from sklearn.feature_extraction.text import CountVectorizer as SkCV
+    ref = SkCV().fit(DOCS)
+    assert cv.vocabulary_ == ref.vocabulary_
+    np.testing.assert_array_equal(m.toarray(), ref.transform(DOCS).toarray())
+
+
+def test_feature_hasher():
+    items = [{"a": 1, "b": 2}, {"b": 3, "c": 1}]
+    fh = FeatureHasher(n_features=32)
+    m = sp.vstack(fh.transform(items).gather())
+    assert m.shape == (2, 32)
+
+
+def test_spectral_clustering():
+    rng = np.random.default_rng(2)
+    k = 3
+    centers = np.array([[0, 0], [8, 8], [-8, 8]])
+    lab = rng.integers(0, k, 900)
+    X = (centers[lab] + rng.standard_normal((900, 2))).astype(np.float32)
+    sc = SpectralClustering(n_clusters=k, n_components=60, random_state=0)
+    got = sc.fit_predict(from_numpy(X)).gather().numpy()
+    # clusters should align with true blobs up to permutation
+    from itertools import permutations
+    best = max(
+        (np.mean(np.array([p[g] for g in got]) == lab)
+         for p in permutations(range(k))))
+    assert best > 0.9
+
+
+def test_utils_svd_flip():
+    from mi355ml.utils import svd_flip
+    u = torch.randn(10, 3)
+    v = torch.randn(3, 5)
+    u2, v2 = svd_flip(u, v)
+    np.testing.assert_allclose((u2 @ v2).numpy(), (u @ v).numpy(),
+                               rtol=1e-5, atol=1e-5)
+
+
+def test_profiler_counters():
+    from mi355ml.diagnostics import Profiler, Counters, device_memory
+    p = Profiler(use_events=False)
+    with p.track("op1"):
+        _ = sum(range(1000))
+    assert "op1" in p.summary()
+    c = Counters()
+    c.inc("samples", 5)
+    assert c.snapshot()["samples"] == 5
+    device_memory()  # no GPU here: just must not raise
