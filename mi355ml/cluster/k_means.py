@@ -107,7 +107,9 @@ class KMeans(BaseEstimator):
                  oversampling_factor: float = 2.0, max_iter: int = 300,
                  tol: float = 1e-4, random_state: int = 0,
                  init_max_iter: Optional[int] = None, copy_x: bool = True,
-                 n_jobs: int = 1, algorithm: str = "full"):
+                 n_jobs: int = 1, algorithm: str = "full",
+                 checkpoint_path: Optional[str] = None,
+                 checkpoint_every: int = 0):
         self.n_clusters = n_clusters
         self.init = init
         self.oversampling_factor = oversampling_factor
@@ -118,6 +120,8 @@ class KMeans(BaseEstimator):
         self.copy_x = copy_x
         self.n_jobs = n_jobs
         self.algorithm = algorithm
+        self.checkpoint_path = checkpoint_path
+        self.checkpoint_every = checkpoint_every
 
     # -- init dispatch ----------------------------------------------------
     def _init_centers(self, X: DistArray) -> torch.Tensor:
@@ -147,15 +151,24 @@ class KMeans(BaseEstimator):
         return float(var.double().mean()) * self.tol
 
     def fit(self, X, y=None):
+        from ..checkpoint import CheckpointManager
         X = check_array(X)
         k = self.n_clusters
-        centers = self._init_centers(X)
-        tol = self._tolerance(X)
         comm = X.comm
+        mgr = CheckpointManager(self.checkpoint_path,
+                                self.checkpoint_every, comm)
+        start_iter = 0
+        resumed = mgr.resume() if self.checkpoint_every else None
+        if resumed is not None:
+            centers = resumed["centers"].to(device=X.device, dtype=X.dtype)
+            start_iter = int(resumed["iteration"])
+        else:
+            centers = self._init_centers(X)
+        tol = self._tolerance(X)
         d = X.ncols
-        n_iter = 0
+        n_iter = start_iter
         inertia = float("nan")
-        for n_iter in range(1, self.max_iter + 1):
+        for n_iter in range(start_iter + 1, self.max_iter + 1):
             labels, sums, counts, it_inertia = kmeans_assign(X.local, centers)
             # ONE fused all-reduce: [sums | counts | inertia]  (C3)
             buf = torch.empty(k * d + k + 1, dtype=torch.float64,
@@ -175,6 +188,8 @@ class KMeans(BaseEstimator):
                 centers.double())
             shift = float(((new_centers - centers.double()) ** 2).sum())
             centers = new_centers.to(X.dtype).contiguous()
+            if self.checkpoint_every:
+                mgr.step(n_iter, centers=centers)
             if shift <= tol:
                 break
         self.cluster_centers_ = centers

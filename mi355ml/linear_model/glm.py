@@ -26,7 +26,8 @@ class _GLMBase(BaseEstimator):
     family: str = "normal"
 
     def __init__(self, penalty="l2", C=1.0, fit_intercept=True,
-                 solver="lbfgs", max_iter=100, tol=1e-4, random_state=None):
+                 solver="lbfgs", max_iter=100, tol=1e-4, random_state=None,
+                 eta0=0.5):
         self.penalty = penalty
         self.C = C
         self.fit_intercept = fit_intercept
@@ -34,6 +35,7 @@ class _GLMBase(BaseEstimator):
         self.max_iter = max_iter
         self.tol = tol
         self.random_state = random_state
+        self.eta0 = eta0
 
     def fit(self, X, y):
         X, y = check_X_y(X, y)
@@ -63,6 +65,33 @@ class _GLMBase(BaseEstimator):
         self.intercept_ = float(b)
         self.n_iter_ = nit
         self.n_features_in_ = X.ncols
+        return self
+
+    def partial_fit(self, X, y, classes=None):
+        """One distributed SGD step on this batch (fused kernel + one
+        allreduce) — the native-estimator analog of sklearn partial_fit,
+        usable under wrappers.Incremental and the Hyperband family."""
+        import math
+        from ..glm.algorithms import _dist_value_grad
+        X, y = check_X_y(X, y)
+        d = X.ncols
+        dev = X.device
+        if not hasattr(self, "coef_") or self.coef_ is None:
+            self.coef_ = torch.zeros(d, dtype=torch.float32, device=dev)
+            self.intercept_ = 0.0
+            self._t = 0
+            self.n_features_in_ = d
+        w = self.coef_.to(dev)
+        loss, gw, gb = _dist_value_grad(X, y, w, self.intercept_,
+                                        self.family)
+        n = max(X.global_rows, 1)
+        lam = (1.0 / self.C) if (self.C and self.penalty == "l2") else 0.0
+        self._t += 1
+        eta = self.eta0 / math.sqrt(self._t)
+        step = gw / n + lam * w.double() / n
+        self.coef_ = (w.double() - eta * step).float()
+        if self.fit_intercept:
+            self.intercept_ = float(self.intercept_ - eta * gb / n)
         return self
 
     def _linear(self, X) -> DistArray:

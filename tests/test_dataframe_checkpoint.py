@@ -1,0 +1,74 @@
+"""Categorizer/DummyEncoder (dataframe partitions) + checkpoint/resume +
+native GLM partial_fit."""
+
+import numpy as np
+import pandas as pd
+import torch
+
+from mi355ml.core.array import from_numpy
+from mi355ml.core.bag import DistList
+from mi355ml.preprocessing import Categorizer, DummyEncoder
+
+
+def _frames():
+    df = pd.DataFrame({
+        "a": ["x", "y", "x", "z", "y", "x"],
+        "b": [1.0, 2.0, 3.0, 4.0, 5.0, 6.0],
+    })
+    return DistList.from_global([df.iloc[:3], df.iloc[3:]])
+
+
+def test_categorizer_dummy_encoder():
+    frames = _frames()
+    cz = Categorizer().fit(frames)
+    assert cz.categories_ == {"a": ["x", "y", "z"]}
+    cat = cz.transform(frames)
+    for f in cat.gather():
+        assert isinstance(f["a"].dtype, pd.CategoricalDtype)
+        assert list(f["a"].cat.categories) == ["x", "y", "z"]
+    de = DummyEncoder().fit(cat)
+    enc = de.transform(cat)
+    out = pd.concat(enc.gather())
+    assert {"a_x", "a_y", "a_z", "b"} <= set(out.columns)
+    assert out.shape == (6, 4)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    from mi355ml.checkpoint import save_checkpoint, load_checkpoint
+    p = str(tmp_path / "s.ckpt")
+    save_checkpoint(p, {"iteration": 5, "centers": torch.ones(3, 2)})
+    st = load_checkpoint(p)
+    assert st["iteration"] == 5
+    np.testing.assert_array_equal(st["centers"].numpy(), np.ones((3, 2)))
+
+
+def test_kmeans_checkpoint_resume(tmp_path):
+    from mi355ml.cluster import KMeans
+    rng = np.random.default_rng(0)
+    centers = rng.uniform(-8, 8, (3, 4))
+    lab = rng.integers(0, 3, 1200)
+    X = (centers[lab] + rng.standard_normal((1200, 4))).astype(np.float32)
+    p = str(tmp_path / "km.ckpt")
+    km1 = KMeans(n_clusters=3, random_state=0, max_iter=3,
+                 checkpoint_path=p, checkpoint_every=1)
+    km1.fit(from_numpy(X))
+    import os
+    assert os.path.exists(p)
+    km2 = KMeans(n_clusters=3, random_state=0, max_iter=50,
+                 checkpoint_path=p, checkpoint_every=1)
+    km2.fit(from_numpy(X))  # resumes from iteration 3
+    assert km2.n_iter_ >= 3
+    assert km2.inertia_ <= km1.inertia_ * 1.001
+
+
+def test_native_glm_partial_fit_incremental():
+    from mi355ml.linear_model import LogisticRegression
+    rng = np.random.default_rng(1)
+    X = rng.standard_normal((2000, 5)).astype(np.float32)
+    w = rng.standard_normal(5) * 2
+    y = ((1 / (1 + np.exp(-(X @ w)))) > rng.uniform(size=2000)).astype(
+        np.float32)
+    lr = LogisticRegression(C=1e3)
+    for _ in range(30):
+        lr.partial_fit(from_numpy(X), from_numpy(y))
+    assert lr.score(from_numpy(X), from_numpy(y)) > 0.85
