@@ -29,6 +29,10 @@ def check_array(X: ArrayLike, dtype: torch.dtype = torch.float32,
     analog, dask_ml/utils.py)."""
     if isinstance(X, DistArray):
         if dtype is not None and X.dtype != dtype:
+            from .core.spill import SpilledDistArray
+            if isinstance(X, SpilledDistArray):
+                return SpilledDistArray(X.local.to(dtype), X.global_rows,
+                                        X.row_offset, X.comm, X.block_rows)
             return X.astype(dtype)
         return X
     if isinstance(X, np.ndarray):

@@ -26,6 +26,36 @@ from ..metrics.pairwise import euclidean_distances
 from ..ops import closest_center, kmeans_assign
 
 
+def _iter_blocks(X: DistArray):
+    """Yield (local_offset, device_block) — one block for resident shards,
+    streamed blocks for spilled shards."""
+    from ..core.spill import SpilledDistArray
+    if isinstance(X, SpilledDistArray):
+        yield from X.device_blocks()
+    else:
+        yield 0, X.local
+
+
+def _shard_assign(X: DistArray, centers: torch.Tensor):
+    """Per-shard (sums, counts, inertia) accumulated over blocks."""
+    sums = counts = inertia = None
+    for _, blk in _iter_blocks(X):
+        _, s, c, i = kmeans_assign(blk, centers.to(blk.device))
+        if sums is None:
+            sums, counts, inertia = s, c, i
+        else:
+            sums, counts, inertia = sums + s, counts + c, inertia + i
+    return sums, counts, inertia
+
+
+def _shard_labels(X: DistArray, centers: torch.Tensor) -> torch.Tensor:
+    outs = []
+    for _, blk in _iter_blocks(X):
+        lab, _ = closest_center(blk, centers.to(blk.device))
+        outs.append(lab.to(X.local.device))
+    return torch.cat(outs) if len(outs) > 1 else outs[0]
+
+
 def _global_row(X: DistArray, gidx: int) -> torch.Tensor:
     """Fetch one global row onto every rank (tiny allreduce)."""
     lo = X.row_offset
@@ -50,9 +80,11 @@ def k_init(X: DistArray, n_clusters: int, random_state: int = 0,
     centers = _global_row(X, first)[None, :]
 
     def _phi(c: torch.Tensor) -> float:
-        _, d2 = closest_center(X.local, c)
-        s = d2.double().sum()
-        t = s.reshape(1)
+        s = 0.0
+        for _, blk in _iter_blocks(X):
+            _, d2 = closest_center(blk, c.to(blk.device))
+            s += float(d2.double().sum())
+        t = torch.tensor([s], dtype=torch.float64, device=dev)
         if comm.world_size > 1:
             comm.all_reduce(t)
         return float(t[0])
@@ -68,12 +100,15 @@ def k_init(X: DistArray, n_clusters: int, random_state: int = 0,
     for _ in range(rounds):
         if phi <= 0.0:
             break
-        _, d2 = closest_center(X.local, centers)
-        p = (ell * d2.double() / phi).clamp(max=1.0)
-        u = torch.rand(p.shape[0], generator=sg, device=dev,
-                       dtype=torch.float64)
-        sel = u < p
-        cand = X.local[sel]
+        cands = []
+        for lo, blk in _iter_blocks(X):
+            _, d2 = closest_center(blk, centers.to(blk.device))
+            p = (ell * d2.double() / phi).clamp(max=1.0)
+            u = torch.rand(p.shape[0], generator=sg, device=dev,
+                           dtype=torch.float64).to(p.device)
+            cands.append(blk[u < p].to(dev))
+        cand = (torch.cat(cands, dim=0) if len(cands) > 1 else cands[0]) \
+            if cands else torch.empty(0, X.ncols, dtype=X.dtype, device=dev)
         parts = comm.all_gather_varlen(cand.contiguous())
         new = torch.cat(parts, dim=0) if comm.world_size > 1 else cand
         if new.shape[0]:
@@ -81,8 +116,9 @@ def k_init(X: DistArray, n_clusters: int, random_state: int = 0,
         phi = _phi(centers)
 
     # weight candidates by assignment counts, then local weighted KMeans
-    labels, _ = closest_center(X.local, centers)
-    w = torch.bincount(labels.long(), minlength=centers.shape[0]).double()
+    labels = _shard_labels(X, centers)
+    w = torch.bincount(labels.long(),
+                       minlength=centers.shape[0]).double().to(dev)
     if comm.world_size > 1:
         comm.all_reduce(w)
     if centers.shape[0] <= k:
@@ -169,7 +205,7 @@ class KMeans(BaseEstimator):
         n_iter = start_iter
         inertia = float("nan")
         for n_iter in range(start_iter + 1, self.max_iter + 1):
-            labels, sums, counts, it_inertia = kmeans_assign(X.local, centers)
+            sums, counts, it_inertia = _shard_assign(X, centers)
             # ONE fused all-reduce: [sums | counts | inertia]  (C3)
             buf = torch.empty(k * d + k + 1, dtype=torch.float64,
                               device=X.device)
@@ -196,7 +232,7 @@ class KMeans(BaseEstimator):
         self.inertia_ = inertia
         self.n_iter_ = n_iter
         self.n_features_in_ = d
-        labels, _ = closest_center(X.local, centers)
+        labels = _shard_labels(X, centers)
         self.labels_ = DistArray(labels.to(torch.int64), X.global_rows,
                                  X.row_offset, comm)
         return self
@@ -207,8 +243,7 @@ class KMeans(BaseEstimator):
     def predict(self, X):
         check_is_fitted(self, "cluster_centers_")
         X = check_array(X)
-        c = self.cluster_centers_.to(X.device)
-        labels, _ = closest_center(X.local, c)
+        labels = _shard_labels(X, self.cluster_centers_.to(X.device))
         return DistArray(labels.to(torch.int64), X.global_rows, X.row_offset,
                          X.comm)
 
@@ -219,8 +254,11 @@ class KMeans(BaseEstimator):
     def score(self, X, y=None):
         check_is_fitted(self, "cluster_centers_")
         X = check_array(X)
-        _, d2 = closest_center(X.local, self.cluster_centers_.to(X.device))
-        t = d2.double().sum().reshape(1)
+        s = 0.0
+        for _, blk in _iter_blocks(X):
+            _, d2 = closest_center(blk, self.cluster_centers_.to(blk.device))
+            s += float(d2.double().sum())
+        t = torch.tensor([s], dtype=torch.float64, device=X.device)
         if X.comm.world_size > 1:
             X.comm.all_reduce(t)
         return -float(t[0])

@@ -27,11 +27,30 @@ from .families import get_family
 from .regularizers import get_regularizer
 
 
+def _shard_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
+                      family: str):
+    """Per-shard fused (loss, gw, gb); streams blocks for spilled shards."""
+    from ..core.spill import SpilledDistArray
+    if isinstance(X, SpilledDistArray):
+        loss = gw = gb = None
+        for lo, blk in X.device_blocks():
+            yb = y.local[lo:lo + blk.shape[0]]
+            if yb.device != blk.device:
+                yb = yb.to(blk.device, non_blocking=True)
+            l_, gw_, gb_ = glm_value_grad(blk, yb, w.to(blk.device),
+                                          torch.tensor(b), family)
+            if loss is None:
+                loss, gw, gb = l_, gw_, gb_
+            else:
+                loss, gw, gb = loss + l_, gw + gw_, gb + gb_
+        return loss, gw, gb
+    return glm_value_grad(X.local, y.local, w, torch.tensor(b), family)
+
+
 def _dist_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
                      family: str) -> Tuple[float, torch.Tensor, float]:
     """Global (loss, grad_w, grad_b) — fused kernel + one fused allreduce."""
-    loss, gw, gb = glm_value_grad(X.local, y.local, w, torch.tensor(b),
-                                  family)
+    loss, gw, gb = _shard_value_grad(X, y, w, b, family)
     buf = torch.empty(2 + gw.shape[0], dtype=torch.float64, device=gw.device)
     buf[0] = loss
     buf[1] = gb
@@ -43,8 +62,7 @@ def _dist_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
 
 def _local_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
                       family: str) -> Tuple[float, torch.Tensor, float]:
-    loss, gw, gb = glm_value_grad(X.local, y.local, w, torch.tensor(b),
-                                  family)
+    loss, gw, gb = _shard_value_grad(X, y, w, b, family)
     return float(loss), gw, float(gb)
 
 

@@ -1,0 +1,57 @@
+"""Host-spill tier: streamed reductions and estimator fits match resident."""
+
+import numpy as np
+import torch
+
+from mi355ml.core.array import from_numpy
+from mi355ml.core.spill import spill_rows, to_spilled
+from mi355ml.cluster import KMeans
+from mi355ml.linear_model import LogisticRegression
+
+
+def _data(n=4000, f=6, seed=0):
+    rng = np.random.default_rng(seed)
+    return rng.standard_normal((n, f)).astype(np.float32)
+
+
+def test_spilled_moments_match():
+    a = _data()
+    X = from_numpy(a)
+    S = spill_rows(torch.from_numpy(a), block_rows=700)
+    n1, m1, v1 = X.moments()
+    n2, m2, v2 = S.moments()
+    assert n1 == n2
+    torch.testing.assert_close(m1, m2, rtol=1e-10, atol=1e-8)
+    torch.testing.assert_close(v1, v2, rtol=1e-8, atol=1e-6)
+    torch.testing.assert_close(X.min(), S.min())
+    torch.testing.assert_close(X.max(), S.max())
+
+
+def test_spilled_logreg_matches_resident():
+    rng = np.random.default_rng(1)
+    a = _data(3000, 5, seed=1)
+    w = rng.standard_normal(5)
+    y = ((1 / (1 + np.exp(-(a @ w)))) > rng.uniform(size=3000)).astype(
+        np.float32)
+    lr1 = LogisticRegression(C=1.0, max_iter=100).fit(from_numpy(a),
+                                                      from_numpy(y))
+    S = spill_rows(torch.from_numpy(a), block_rows=800)
+    Sy = from_numpy(y)
+    lr2 = LogisticRegression(C=1.0, max_iter=100).fit(S, Sy)
+    np.testing.assert_allclose(lr1.coef_.numpy(), lr2.coef_.numpy(),
+                               rtol=1e-4, atol=1e-5)
+
+
+def test_spilled_kmeans():
+    rng = np.random.default_rng(2)
+    centers = rng.uniform(-8, 8, (4, 5))
+    lab = rng.integers(0, 4, 3000)
+    a = (centers[lab] + rng.standard_normal((3000, 5))).astype(np.float32)
+    km1 = KMeans(n_clusters=4, random_state=0).fit(from_numpy(a))
+    S = spill_rows(torch.from_numpy(a), block_rows=640)
+    km2 = KMeans(n_clusters=4, random_state=0).fit(S)
+    # same seeded init + same data: same result
+    np.testing.assert_allclose(km1.cluster_centers_.numpy(),
+                               km2.cluster_centers_.numpy(), rtol=1e-4,
+                               atol=1e-4)
+    assert abs(km1.inertia_ - km2.inertia_) / km1.inertia_ < 1e-4
