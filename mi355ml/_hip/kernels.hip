@@ -254,18 +254,26 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
       dot += __shfl_xor(dot, 1, 4);
       dot += __shfl_xor(dot, 2, 4);
       if (r < rows && q == 0) {
-        const double z = (double)(dot + b);
-        const double yv = (double)yt[r];
         double li, ri;
         if (FAM == 0) {
-          const double sp = z > 0.0 ? z + log1p(exp(-z)) : log1p(exp(z));
-          li = sp - yv * z;
-          ri = 1.0 / (1.0 + exp(-z)) - yv;
+          // f32 link: f64 exp/log1p are software-emulated on CDNA4 and
+          // dominated the z-phase; per-row f32 error (~1e-7 rel) vanishes
+          // in the f64 partial sums.
+          const float z = dot + b;
+          const float yv = yt[r];
+          const float sp = z > 0.0f ? z + log1pf(__expf(-z))
+                                    : log1pf(__expf(z));
+          li = (double)(sp - yv * z);
+          ri = (double)(1.0f / (1.0f + __expf(-z)) - yv);
         } else if (FAM == 1) {
+          const double z = (double)(dot + b);
+          const double yv = (double)yt[r];
           const double df = z - yv;
           li = 0.5 * df * df;
           ri = df;
         } else {
+          const double z = (double)(dot + b);
+          const double yv = (double)yt[r];
           const double ez = exp(z);
           li = ez - yv * z;
           ri = ez - yv;
@@ -481,25 +489,25 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       }
       int k0 = 0;
       for (; k0 + 8 <= d4; k0 += 8) {
+        // both steps' MFMAs back to back, then BOTH refills adjacent so the
+        // per-tile (kn0, kn1) reads merge into ds_read2_b32 (halves the
+        // LDS instruction count; the refills get 14 MFMA issues of slack)
 #pragma unroll
         for (int t = 0; t < NTILES; ++t)
           acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],
                                                         0, 0, 0);
-        {
-          const int kn = min(k0 + 8, kl);
-          a0 = xrow[kn];
-#pragma unroll
-          for (int t = 0; t < NTILES; ++t) b0[t] = cb[16 * t * d4 + kn];
-        }
 #pragma unroll
         for (int t = 0; t < NTILES; ++t)
           acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1[t], acc[t],
                                                         0, 0, 0);
-        {
-          const int kn = min(k0 + 12, kl);
-          a1 = xrow[kn];
+        const int kn0 = min(k0 + 8, kl);
+        const int kn1 = min(k0 + 12, kl);
+        a0 = xrow[kn0];
+        a1 = xrow[kn1];
 #pragma unroll
-          for (int t = 0; t < NTILES; ++t) b1[t] = cb[16 * t * d4 + kn];
+        for (int t = 0; t < NTILES; ++t) {
+          b0[t] = cb[16 * t * d4 + kn0];
+          b1[t] = cb[16 * t * d4 + kn1];
         }
       }
       if (k0 < d4) {  // d4 % 8 == 4 tail: one more step (data in buf0)

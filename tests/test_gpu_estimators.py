@@ -82,3 +82,36 @@ def test_kmeans_gpu_matches_cpu_one_iter():
     new_g = sums_g.cpu() / cnt_g.cpu().clamp(min=1)[:, None]
     torch.testing.assert_close(new_g, new_c, rtol=1e-5, atol=1e-5)
     torch.testing.assert_close(in_g.cpu(), in_c, rtol=1e-5, atol=1e-2)
+
+
+def test_spill_gpu_streams_blocks():
+    import torch
+    from mi355ml.core.spill import spill_rows
+    from mi355ml.core.array import shard_rows
+    from mi355ml.linear_model import LogisticRegression
+    g = torch.Generator().manual_seed(11)
+    a = torch.randn(2_000_000, 50, generator=g)
+    w = torch.randn(50, generator=g)
+    y = (torch.sigmoid(a @ w) > torch.rand(2_000_000, generator=g)).float()
+    S = spill_rows(a, block_rows=500_000)   # host-pinned shard
+    assert S.local.device.type == "cpu" and S.device.type == "cuda"
+    lr = LogisticRegression(C=1.0, max_iter=50).fit(S, shard_rows(y))
+    lr2 = LogisticRegression(C=1.0, max_iter=50).fit(shard_rows(a),
+                                                     shard_rows(y))
+    torch.testing.assert_close(lr.coef_.cpu(), lr2.coef_.cpu(), rtol=1e-3,
+                               atol=1e-4)
+
+
+def test_kmeans_gpu_bitwise_deterministic():
+    """Run-to-run determinism: no atomics in the fused path."""
+    from mi355ml.ops import kmeans_assign
+    import torch
+    g = torch.Generator().manual_seed(13)
+    x = torch.randn(1_000_000, 50, generator=g).cuda()
+    c = torch.randn(100, 50, generator=g).cuda()
+    l1, s1, c1, i1 = kmeans_assign(x, c)
+    l2, s2, c2, i2 = kmeans_assign(x, c)
+    assert torch.equal(l1, l2)
+    assert torch.equal(s1, s2)
+    assert torch.equal(c1, c2)
+    assert torch.equal(i1, i2)
