@@ -22,6 +22,8 @@ void launch_affine_transform(const float*, const float*, const float*,
                              float*, long, int, int, hipStream_t);
 void launch_kmeans_probe(const float*, const float*, long, int, int, int*,
                          float*, float*, double*, int, int, hipStream_t);
+void launch_glm_probe(const float*, const float*, const float*, float, long,
+                      int, int, double*, double*, double*, int, hipStream_t);
 size_t kmeans_smem_bytes(int k, int d, int update);
 }
 
@@ -151,6 +153,23 @@ std::vector<torch::Tensor> closest_center(torch::Tensor x,
   return {labels, dists};
 }
 
+torch::Tensor glm_phase_probe(torch::Tensor x, torch::Tensor y,
+                              torch::Tensor w, long phases) {
+  const long m = x.size(0);
+  const int d = (int)x.size(1);
+  const long tiles = (m + 63) / 64;
+  const int nb = (int)std::max(1L, std::min((long)kMaxBlocks, tiles));
+  auto opts = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
+  auto loss_slab = torch::zeros({nb}, opts);
+  auto gw_slab = torch::zeros({nb, d}, opts);
+  auto gb_slab = torch::zeros({nb}, opts);
+  launch_glm_probe(x.data_ptr<float>(), y.data_ptr<float>(),
+                   w.data_ptr<float>(), 0.1f, m, d, (int)phases,
+                   loss_slab.data_ptr<double>(), gw_slab.data_ptr<double>(),
+                   gb_slab.data_ptr<double>(), nb, cur_stream());
+  return loss_slab;
+}
+
 torch::Tensor kmeans_phase_probe(torch::Tensor x, torch::Tensor centers,
                                  long phases) {
   const long m = x.size(0);
@@ -203,6 +222,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused argmin-distance labels + squared distances");
   mod.def("affine_transform", &affine_transform,
           "fused per-column affine transform");
+  mod.def("glm_phase_probe", &glm_phase_probe,
+          "glm ablation (bit0 stage, bit1 z, bit2 grad)");
   mod.def("kmeans_phase_probe", &kmeans_phase_probe,
           "phase-mask ablation probe (bit0 stage, bit1 compute, bit2 update)");
 }

@@ -17,7 +17,8 @@
 #include <math.h>
 
 #define NT 256      // threads per block
-#define GLM_TR 64   // GLM row-tile height
+#define GLM_TR 64     // GLM row-tile height
+#define GLM_CSLOTS 8  // grad column slots per lane (d <= 512)
 
 // ---------------------------------------------------------------------------
 // K7/K8: column moments (n, sum, sumsq->M2) — one pass, per-WG partials.
@@ -162,7 +163,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 // LDS carve (all dynamic, 16B-aligned base): w[d] | xtile[TR*ld] | y[TR] |
 // rres[TR] | red[2*NT doubles].
 // ---------------------------------------------------------------------------
-template <int FAM>
+template <int FAM, int PHASES = 7>
 __global__ __launch_bounds__(NT) void glm_value_grad_k(
     const float* __restrict__ x, const float* __restrict__ y,
     const float* __restrict__ w, float b, long m, int d, int ld, int nbuf,
@@ -182,13 +183,13 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
   for (int i = tid; i < d; i += NT) sw[i] = w[i];
 
   double lacc = 0.0, gbacc = 0.0;
-  double gacc0 = 0.0, gacc1 = 0.0;  // column slots tid and tid+NT
+  double gacc[GLM_CSLOTS] = {};  // per-wave column slots lane + 64*sl
 
   const long tiles = (m + GLM_TR - 1) / GLM_TR;
   // Double-buffered LDS-DMA staging (same pattern as the kmeans kernel):
   // tile t+1 streams into buffer cur^1 by global_load_lds while tile t is
   // consumed; raw barriers so the DMA stays in flight across them.
-  const bool use_glds = ((d & 3) == 0) && (ld == d);
+  const bool use_glds = ((d & 3) == 0) && (ld == d) && ((PHASES & 1) != 0);
   const bool dbuf = (nbuf == 2);
   int cur = 0;
   long tile = blockIdx.x;
@@ -222,7 +223,7 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
     }
   };
 
-  if (tile < tiles) {
+  if ((PHASES & 1) && tile < tiles) {
     if (use_glds && m - tile * GLM_TR >= GLM_TR) {
       stage_glds(xtbuf, ytbuf, tile);
       vm_drain();
@@ -242,10 +243,12 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
     const int nxt = dbuf ? (cur ^ 1) : 0;
     if (pf)
       stage_glds(xtbuf + nxt * GLM_TR * ld, ytbuf + nxt * GLM_TR, ntile);
-    // z-phase: 4 lanes per row
-    {
-      const int r = tid >> 2;
-      const int q = tid & 3;
+    // z-phase: WAVE-LOCAL (wave w owns rows [16w, 16w+16)) so the z->grad
+    // dependency never crosses waves and the intra-tile barrier disappears
+    const int wrow0 = wave * 16;
+    if (PHASES & 2) {
+      const int r = wrow0 + (lane >> 2);
+      const int q = lane & 3;
       float dot = 0.0f;
       if (r < rows) {
         const float* xrow = xt + r * ld;
@@ -283,25 +286,27 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
         rres[r] = (float)ri;
       }
     }
-    lds_barrier();
-    // grad-phase: thread owns columns tid and tid+NT
-    if (tid < d) {
-      double acc = 0.0;
-      const float* xp = xt + tid;
-      for (int rr = 0; rr < rows; ++rr)
-        acc += (double)rres[rr] * (double)xp[rr * ld];
-      gacc0 += acc;
-    }
-    if (tid + NT < d) {
-      double acc = 0.0;
-      const float* xp = xt + tid + NT;
-      for (int rr = 0; rr < rows; ++rr)
-        acc += (double)rres[rr] * (double)xp[rr * ld];
-      gacc1 += acc;
+    // grad-phase: wave-local rank-1 accumulation — lane owns columns
+    // lane, lane+64, ... over THIS wave's 16 rows (partials combined across
+    // waves once at kernel end). No barrier between z and grad: the LDS
+    // dependency is within the wave.
+    if (PHASES & 4) {
+      const int rend = min(16, rows - wrow0);
+#pragma unroll
+      for (int sl = 0; sl < GLM_CSLOTS; ++sl) {
+        const int c = lane + 64 * sl;
+        if (c < d) {
+          double acc = 0.0;
+          const float* xp = xt + wrow0 * ld + c;
+          for (int rr = 0; rr < rend; ++rr)
+            acc += (double)rres[wrow0 + rr] * (double)xp[rr * ld];
+          gacc[sl] += acc;
+        }
+      }
     }
     if (pf) {
       vm_drain();  // next tile's DMA has the whole compute to land
-    } else if (ntile < tiles) {
+    } else if ((PHASES & 1) && ntile < tiles) {
       stage_direct(xtbuf + nxt * GLM_TR * ld, ytbuf + nxt * GLM_TR, ntile);
     }
     lds_barrier();
@@ -322,8 +327,17 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
     loss_slab[blockIdx.x] = red[0];
     gb_slab[blockIdx.x] = red[NT];
   }
-  if (tid < d) gw_slab[(long)blockIdx.x * d + tid] = gacc0;
-  if (tid + NT < d) gw_slab[(long)blockIdx.x * d + tid + NT] = gacc1;
+  // combine the 4 waves' gradient partials through LDS (xtbuf reusable now)
+  double* gred = (double*)xtbuf;  // [4 * d] doubles (d <= 512 -> 16 KB)
+#pragma unroll
+  for (int sl = 0; sl < GLM_CSLOTS; ++sl) {
+    const int c = lane + 64 * sl;
+    if (c < d) gred[wave * d + c] = gacc[sl];
+  }
+  __syncthreads();
+  for (int c = tid; c < d; c += NT)
+    gw_slab[(long)blockIdx.x * d + c] =
+        gred[c] + gred[d + c] + gred[2 * d + c] + gred[3 * d + c];
 }
 
 // ---------------------------------------------------------------------------
@@ -745,6 +759,24 @@ void launch_kmeans_assign(const float* x, const float* cen, long m, int d,
                           hipStream_t stream) {
   kmeans_dispatch<1>(x, cen, m, d, k, labels, nullptr, sums_slab, cnt_slab,
                      inertia_slab, nblocks, stream);
+}
+
+void launch_glm_probe(const float* x, const float* y, const float* w,
+                      float b, long m, int d, int phases, double* loss_slab,
+                      double* gw_slab, double* gb_slab, int nblocks,
+                      hipStream_t stream) {
+  const int ld = ((d & 3) == 0) ? d : (d | 1);
+  const size_t smem =
+      ((((((d + 3) & ~3) + 2 * GLM_TR * ld + 3 * GLM_TR) * sizeof(float) +
+         15) & ~15)) + 2 * NT * sizeof(double);
+#define GP(PH)                                                              \
+  case PH:                                                                  \
+    hipLaunchKernelGGL((glm_value_grad_k<0, PH>), dim3(nblocks), dim3(NT),  \
+                       smem, stream, x, y, w, b, m, d, ld, 2, loss_slab,    \
+                       gw_slab, gb_slab);                                   \
+    break
+  switch (phases) { GP(0); GP(1); GP(3); GP(5); GP(7); }
+#undef GP
 }
 
 void launch_kmeans_probe(const float* x, const float* cen, long m, int d,

@@ -33,7 +33,7 @@ class PCA(BaseEstimator, TransformerMixin):
             return self.svd_solver
         return "randomized" if k < 0.8 * d else "full"
 
-    def _fit(self, X: DistArray):
+    def _fit(self, X: DistArray, need_u: bool = False):
         d = X.ncols
         n = X.global_rows
         k = self.n_components if self.n_components is not None else min(n, d)
@@ -42,18 +42,16 @@ class PCA(BaseEstimator, TransformerMixin):
         Xc = X.map_local(lambda t: affine_transform(t, ones, -mean))
         solver = self._solver(d, k)
         if solver in ("full", "tsqr"):
-            U, s, Vh = tsqr(Xc, compute_svd=True)
-            Uk = DistArray(U._as2d()[:, :k], X.global_rows, X.row_offset,
-                           X.comm)
-            s_all = s
+            U, s, Vh = tsqr(Xc, compute_svd=True, compute_q=need_u)
+            Uk = (DistArray(U._as2d()[:, :k], X.global_rows, X.row_offset,
+                            X.comm) if need_u else None)
             s = s[:k]
             Vh = Vh[:k]
         elif solver == "randomized":
             U, s, Vh = svd_compressed(
                 Xc, k, n_power_iter=self.iterated_power,
-                seed=int(self.random_state or 0))
+                seed=int(self.random_state or 0), compute_u=need_u)
             Uk = U
-            s_all = None
         else:
             raise ValueError(f"unknown svd_solver {solver!r}")
         n_f = float(n)
@@ -84,7 +82,7 @@ class PCA(BaseEstimator, TransformerMixin):
 
     def fit_transform(self, X, y=None):
         X = check_array(X)
-        U, s, _ = self._fit(X)
+        U, s, _ = self._fit(X, need_u=True)
         if self.whiten:
             import math
             scale = math.sqrt(max(X.global_rows - 1, 1))

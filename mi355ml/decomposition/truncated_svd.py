@@ -22,17 +22,19 @@ class TruncatedSVD(BaseEstimator, TransformerMixin):
         self.random_state = random_state
         self.tol = tol
 
-    def _fit(self, X: DistArray):
+    def _fit(self, X: DistArray, need_u: bool = False):
         k = self.n_components
         if self.algorithm in ("tsqr", "full"):
-            U, s, Vh = tsqr(X, compute_svd=True)
-            U = DistArray(U._as2d()[:, :k], X.global_rows, X.row_offset,
-                          X.comm)
+            U, s, Vh = tsqr(X, compute_svd=True, compute_q=need_u)
+            if need_u:
+                U = DistArray(U._as2d()[:, :k], X.global_rows, X.row_offset,
+                              X.comm)
             s = s[:k]
             Vh = Vh[:k]
         elif self.algorithm == "randomized":
             U, s, Vh = svd_compressed(X, k, n_power_iter=self.n_iter,
-                                      seed=int(self.random_state or 0))
+                                      seed=int(self.random_state or 0),
+                                      compute_u=need_u)
         else:
             raise ValueError(f"unknown algorithm {self.algorithm!r}")
         n = float(X.global_rows)
@@ -53,7 +55,7 @@ class TruncatedSVD(BaseEstimator, TransformerMixin):
 
     def fit_transform(self, X, y=None):
         X = check_array(X)
-        U, s, _ = self._fit(X)
+        U, s, _ = self._fit(X, need_u=True)
         return U.map_local(lambda t: t * s.to(t.dtype))
 
     def transform(self, X):

@@ -1,15 +1,20 @@
 """Distributed tall-skinny linear algebra (K10–K12, C5–C6 in SURVEY §2.7).
 
 tsqr: the reference's recursive blockwise TSQR graph
-(dask/array/linalg.py:52, algorithm in SURVEY A.1) collapses on one node:
-per-rank panel QR (rocSOLVER via torch.linalg.qr), ONE all-gather of the
-W x [d, d] R panels (C5), a redundant-deterministic QR/SVD of the stacked R
-on every rank (no broadcast needed), and a local Q1 @ Q2 GEMM.
+(dask/array/linalg.py:52, algorithm in SURVEY A.1) mapped to one node:
+stage-1 panel QRs run per ROW BLOCK of the local shard (bounded blocks keep
+rocSOLVER inside int32 index space and bound the float64 staging memory —
+this is exactly the reference's recursion step A.1-2), the per-rank R
+stacks reduce locally, then ONE all-gather of the W x [d, d] R panels (C5),
+a redundant-deterministic QR/SVD of the stacked R on every rank (no
+broadcast needed), and local Q1 @ Q2 GEMMs for Q only when the caller needs
+it (PCA/TSVD fit paths don't).
 
 svd_compressed: Halko randomized range finder
-(dask/array/linalg.py:616-740): shared Ω from the master seed, per-rank
-GEMMs, TSQR re-orthogonalization per power iteration, allreduce for the
-XᵀQ projections (C6).
+(dask/array/linalg.py:616-740): shared Ω from the master seed, per-rank f32
+GEMMs (the small projected factors are refined in f64), TSQR
+re-orthogonalization per power iteration, allreduce for the XᵀQ
+projections (C6).
 """
 
 from __future__ import annotations
@@ -22,39 +27,81 @@ from ..core.array import DistArray
 from ..core import random as mlrand
 
 
+def _block_rows_for(d: int) -> int:
+    """Stage-1 row-block size: keep the f64 staging block <= ~4 GB and the
+    element count well under 2^31 (rocSOLVER int32 indexing)."""
+    return max(d, min(4_000_000, (1 << 30) // max(d, 1)))
+
+
 def _local_qr(a: torch.Tensor, d: int) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Reduced QR of a local shard, zero-padding shards with < d rows so R is
-    always [d, d] (QR of [A; 0] has the same R; Q rows for the padding are
-    dropped)."""
+    """Reduced QR of one block in the INPUT dtype (the reference's
+    np.linalg.qr also runs in the chunk dtype); the small stacked-R stages
+    upcast to float64. Blocks with < d rows are zero-padded so R is always
+    [d, d]."""
     m = a.shape[0]
     if m >= d:
-        q, r = torch.linalg.qr(a, mode="reduced")
-        return q, r
+        return torch.linalg.qr(a, mode="reduced")
     pad = torch.zeros(d - m, d, dtype=a.dtype, device=a.device)
     q, r = torch.linalg.qr(torch.cat([a, pad], dim=0), mode="reduced")
     return q[:m], r
 
 
-def tsqr(X: DistArray, compute_svd: bool = False):
+def _shard_tsqr_r(a: torch.Tensor, d: int, keep_q: bool):
+    """Blocked stage-1 TSQR of the local shard.
+
+    Returns (blocks, r_local): blocks is a list of
+    (row_lo, Q1_block f32 or None, R1 [d,d] f64) and r_local the [d, d]
+    float64 R of the whole shard; when the shard needed several blocks an
+    inner stacked QR combines them (A.1 recursion, depth 1 suffices)."""
+    m = a.shape[0]
+    br = _block_rows_for(d)
+    blocks = []
+    rs = []
+    for lo in range(0, max(m, 1), br):
+        blk = a[lo:lo + br]
+        q, r = _local_qr(blk, d)
+        blocks.append([lo, q if keep_q else None, r.to(torch.float64)])
+        rs.append(r.to(torch.float64))
+        del q
+    if len(rs) == 1:
+        return blocks, rs[0], None
+    stacked = torch.cat(rs, dim=0)  # [nb*d, d] f64 (small)
+    q2, r = torch.linalg.qr(stacked, mode="reduced")
+    return blocks, r, q2  # q2: [nb*d, d]
+
+
+def tsqr(X: DistArray, compute_svd: bool = False, compute_q: bool = True):
     """QR (or SVD) of a tall-skinny row-sharded matrix.
 
-    Returns (Q DistArray, R tensor) or (U DistArray, s, Vh) with
-    compute_svd=True. R/s/Vh are replicated on every rank.
+    Returns (Q DistArray, R) / (U DistArray, s, Vh); with compute_q=False
+    the Q/U factor is skipped (None) — R/s/Vh are replicated and exact.
     """
     a = X._as2d()
     d = a.shape[1]
     comm = X.comm
-    # float64 panels for numerical stability of the stacked stage
-    q1, r1 = _local_qr(a.to(torch.float64), d)
+    blocks, r_local, q2_inner = _shard_tsqr_r(a, d, keep_q=compute_q)
     if comm.world_size > 1:
-        rs = comm.all_gather(r1.contiguous())
+        rs = comm.all_gather(r_local.contiguous())
         stacked = torch.cat(rs, dim=0)  # [W*d, d]
     else:
-        stacked = r1
-    q2, r = torch.linalg.qr(stacked, mode="reduced")  # replicated determin.
-    q2_own = q2[comm.rank * d:(comm.rank + 1) * d]
+        stacked = r_local
+    q3, r = torch.linalg.qr(stacked, mode="reduced")  # replicated determin.
+    q3_own = q3[comm.rank * d:(comm.rank + 1) * d]  # this rank's block of Q3
+
+    def _assemble(right: torch.Tensor) -> torch.Tensor:
+        """Q_local @ right for the factored Q_local (blocked backprop)."""
+        out = torch.empty(a.shape[0], right.shape[1], dtype=X.dtype,
+                          device=a.device)
+        for i, (lo, q1, _) in enumerate(blocks):
+            piece = right if q2_inner is None else \
+                (q2_inner[i * d:(i + 1) * d] @ right)
+            out[lo:lo + q1.shape[0]] = q1 @ piece.to(X.dtype)
+        return out
+
     if not compute_svd:
-        q = (q1 @ q2_own).to(X.dtype)
+        if not compute_q:
+            return None, r.to(X.dtype)
+        q = _assemble(q3_own)
         return (DistArray(q, X.global_rows, X.row_offset, comm),
                 r.to(X.dtype))
     u_r, s, vh = torch.linalg.svd(r, full_matrices=False)
@@ -64,37 +111,37 @@ def tsqr(X: DistArray, compute_svd: bool = False):
     signs[signs == 0] = 1.0
     vh = vh * signs[:, None]
     u_r = u_r * signs[None, :]
-    u = (q1 @ (q2_own @ u_r)).to(X.dtype)
+    if not compute_q:
+        return None, s.to(X.dtype), vh.to(X.dtype)
+    u = _assemble(q3_own @ u_r)
     return (DistArray(u, X.global_rows, X.row_offset, comm),
             s.to(X.dtype), vh.to(X.dtype))
 
 
 def svd_compressed(X: DistArray, k: int, n_power_iter: int = 2,
-                   n_oversamples: int = 10, seed: int = 0):
-    """Randomized truncated SVD (Halko/Martinsson/Tropp).
+                   n_oversamples: int = 10, seed: int = 0,
+                   compute_u: bool = True):
+    """Randomized truncated SVD (Halko/Martinsson/Tropp), f32 GEMMs.
 
-    Returns (U DistArray [m, k], s [k], Vh [k, d]) replicated s/Vh.
+    Returns (U DistArray [m, k] or None, s [k], Vh [k, d]).
     """
     a = X._as2d()
     m, d = X.global_rows, a.shape[1]
     comm = X.comm
     ell = min(k + n_oversamples, d, m)
     g = mlrand.generator(seed, shard_id=0xABCD, device=a.device)
-    omega = torch.randn(d, ell, generator=g, device=a.device,
-                        dtype=torch.float64)
-    ad = a.to(torch.float64)
-    y = ad @ omega  # [m_local, ell]
-    Y = DistArray(y, m, X.row_offset, comm)
-    Q, _ = tsqr(Y.astype(torch.float64))
+    omega = torch.randn(d, ell, generator=g, device=a.device, dtype=X.dtype)
+    Y = DistArray(a @ omega, m, X.row_offset, comm)
+    Q, _ = tsqr(Y)
     for _ in range(n_power_iter):
         # Z = Xᵀ Q  (allreduce, C6), then re-orthogonalize X·Z
-        z = ad.t() @ Q._as2d().to(torch.float64)
+        z = a.t() @ Q._as2d()
         if comm.world_size > 1:
             comm.all_reduce(z)
-        Y = DistArray(ad @ z, m, X.row_offset, comm)
+        Y = DistArray(a @ z, m, X.row_offset, comm)
         Q, _ = tsqr(Y)
-    # B = Qᵀ X : [ell, d] via allreduce
-    b = Q._as2d().to(torch.float64).t() @ ad
+    # B = Qᵀ X : [ell, d] via allreduce (f64 for the small final factor)
+    b = (Q._as2d().t() @ a).to(torch.float64)
     if comm.world_size > 1:
         comm.all_reduce(b)
     u_b, s, vh = torch.linalg.svd(b, full_matrices=False)
@@ -103,6 +150,8 @@ def svd_compressed(X: DistArray, k: int, n_power_iter: int = 2,
     signs[signs == 0] = 1.0
     vh = vh * signs[:, None]
     u_b = u_b * signs[None, :]
-    u = (Q._as2d().to(torch.float64) @ u_b[:, :k]).to(X.dtype)
-    return (DistArray(u, m, X.row_offset, comm),
-            s[:k].to(X.dtype), vh[:k].to(X.dtype))
+    u = None
+    if compute_u:
+        u = DistArray(Q._as2d() @ u_b[:, :k].to(X.dtype), m, X.row_offset,
+                      comm)
+    return u, s[:k].to(X.dtype), vh[:k].to(X.dtype)

@@ -54,11 +54,11 @@ def test_joblib_roundtrip_fitted_estimators(tmp_path):
         joblib.dump(est, p)
         est2 = joblib.load(p)
         if hasattr(est, "transform"):
-            a = est.transform(from_numpy(X)).gather().numpy()
-            b = est2.transform(from_numpy(X)).gather().numpy()
+            a = est.transform(from_numpy(X)).gather().cpu().numpy()
+            b = est2.transform(from_numpy(X)).gather().cpu().numpy()
         else:
-            a = est.predict(from_numpy(X)).gather().numpy()
-            b = est2.predict(from_numpy(X)).gather().numpy()
+            a = est.predict(from_numpy(X)).gather().cpu().numpy()
+            b = est2.predict(from_numpy(X)).gather().cpu().numpy()
         np.testing.assert_allclose(a, b, rtol=1e-6, atol=1e-6)
 
 

@@ -17,22 +17,22 @@ def _data(n=1000, f=7, seed=0):
 def test_moments_single():
     a = _data()
     X = from_numpy(a)
-    np.testing.assert_allclose(X.mean().numpy(), a.mean(0), rtol=1e-6,
+    np.testing.assert_allclose(X.mean().cpu().numpy(), a.mean(0), rtol=1e-6,
                                atol=1e-6)
-    np.testing.assert_allclose(X.var().numpy(), a.var(0), rtol=1e-5,
+    np.testing.assert_allclose(X.var().cpu().numpy(), a.var(0), rtol=1e-5,
                                atol=1e-6)
-    np.testing.assert_allclose(X.var(ddof=1).numpy(), a.var(0, ddof=1),
+    np.testing.assert_allclose(X.var(ddof=1).cpu().numpy(), a.var(0, ddof=1),
                                rtol=1e-5, atol=1e-6)
-    np.testing.assert_allclose(X.sum().numpy(), a.sum(0, dtype=np.float64),
+    np.testing.assert_allclose(X.sum().cpu().numpy(), a.sum(0, dtype=np.float64),
                                rtol=1e-6)
 
 
 def test_minmax_gather():
     a = _data()
     X = from_numpy(a)
-    np.testing.assert_allclose(X.min().numpy(), a.min(0))
-    np.testing.assert_allclose(X.max().numpy(), a.max(0))
-    np.testing.assert_allclose(X.gather().numpy(), a)
+    np.testing.assert_allclose(X.min().cpu().numpy(), a.min(0))
+    np.testing.assert_allclose(X.max().cpu().numpy(), a.max(0))
+    np.testing.assert_allclose(X.gather().cpu().numpy(), a)
 
 
 def test_dot_xtdot():
@@ -40,11 +40,11 @@ def test_dot_xtdot():
     X = from_numpy(a)
     w = np.random.default_rng(1).standard_normal(a.shape[1]).astype(np.float32)
     z = X.dot(torch.from_numpy(w))
-    np.testing.assert_allclose(z.gather().numpy(), a @ w, rtol=1e-4,
+    np.testing.assert_allclose(z.gather().cpu().numpy(), a @ w, rtol=1e-4,
                                atol=1e-4)
     v = from_numpy((a @ w).astype(np.float32))
     g = X.xt_dot(v)
-    np.testing.assert_allclose(g.numpy()[:, 0],
+    np.testing.assert_allclose(g.cpu().numpy()[:, 0],
                                a.astype(np.float64).T @ (a @ w).astype(np.float64),
                                rtol=1e-5)
 
@@ -62,12 +62,12 @@ def _dist_moments_body(rank, world, a_bytes, shape):
     a = np.frombuffer(a_bytes, dtype=np.float32).reshape(shape).copy()
     X = from_numpy(a)
     assert X.local.shape[0] < a.shape[0]
-    np.testing.assert_allclose(X.mean().numpy(), a.mean(0), rtol=1e-6,
+    np.testing.assert_allclose(X.mean().cpu().numpy(), a.mean(0), rtol=1e-6,
                                atol=1e-6)
-    np.testing.assert_allclose(X.var().numpy(), a.var(0), rtol=1e-5,
+    np.testing.assert_allclose(X.var().cpu().numpy(), a.var(0), rtol=1e-5,
                                atol=1e-6)
-    np.testing.assert_allclose(X.min().numpy(), a.min(0))
-    np.testing.assert_allclose(X.gather().numpy(), a)
+    np.testing.assert_allclose(X.min().cpu().numpy(), a.min(0))
+    np.testing.assert_allclose(X.gather().cpu().numpy(), a)
 
 
 def test_dist_moments_world2(mp_runner):

@@ -39,7 +39,7 @@ def test_checkpoint_roundtrip(tmp_path):
     save_checkpoint(p, {"iteration": 5, "centers": torch.ones(3, 2)})
     st = load_checkpoint(p)
     assert st["iteration"] == 5
-    np.testing.assert_array_equal(st["centers"].numpy(), np.ones((3, 2)))
+    np.testing.assert_array_equal(st["centers"].cpu().numpy(), np.ones((3, 2)))
 
 
 def test_kmeans_checkpoint_resume(tmp_path):

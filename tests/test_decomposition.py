@@ -29,13 +29,13 @@ def test_pca_matches_sklearn(solver):
     X = _lowrank()
     ours = PCA(n_components=5, svd_solver=solver).fit(from_numpy(X))
     ref = SkPCA(n_components=5, svd_solver="full").fit(X)
-    np.testing.assert_allclose(ours.mean_.numpy(), ref.mean_, rtol=1e-4,
+    np.testing.assert_allclose(ours.mean_.cpu().numpy(), ref.mean_, rtol=1e-4,
                                atol=1e-4)
-    np.testing.assert_allclose(ours.singular_values_.numpy(),
+    np.testing.assert_allclose(ours.singular_values_.cpu().numpy(),
                                ref.singular_values_, rtol=1e-3)
-    got = _align_signs(ours.components_.numpy(), ref.components_)
+    got = _align_signs(ours.components_.cpu().numpy(), ref.components_)
     np.testing.assert_allclose(got, ref.components_, rtol=5e-2, atol=5e-3)
-    np.testing.assert_allclose(ours.explained_variance_ratio_.numpy(),
+    np.testing.assert_allclose(ours.explained_variance_ratio_.cpu().numpy(),
                                ref.explained_variance_ratio_, rtol=1e-2,
                                atol=1e-4)
 
@@ -45,12 +45,12 @@ def test_pca_transform_roundtrip():
     pca = PCA(n_components=5).fit(from_numpy(X))
     Z = pca.transform(from_numpy(X))
     assert Z.shape == (X.shape[0], 5)
-    back = pca.inverse_transform(Z).gather().numpy()
+    back = pca.inverse_transform(Z).gather().cpu().numpy()
     # low-rank + small noise: reconstruction error should be tiny
     err = np.abs(back - X).mean() / np.abs(X).mean()
     assert err < 0.02
-    ft = pca.fit_transform(from_numpy(X)).gather().numpy()
-    np.testing.assert_allclose(np.abs(ft), np.abs(Z.gather().numpy()),
+    ft = pca.fit_transform(from_numpy(X)).gather().cpu().numpy()
+    np.testing.assert_allclose(np.abs(ft), np.abs(Z.gather().cpu().numpy()),
                                rtol=1e-2, atol=1e-2)
 
 
@@ -58,9 +58,9 @@ def test_tsvd_matches_sklearn():
     X = _lowrank(seed=2)
     ours = TruncatedSVD(n_components=4, algorithm="tsqr").fit(from_numpy(X))
     ref = SkTSVD(n_components=4, algorithm="arpack").fit(X.astype(np.float64))
-    np.testing.assert_allclose(ours.singular_values_.numpy(),
+    np.testing.assert_allclose(ours.singular_values_.cpu().numpy(),
                                ref.singular_values_, rtol=1e-3)
-    got = _align_signs(ours.components_.numpy(), ref.components_)
+    got = _align_signs(ours.components_.cpu().numpy(), ref.components_)
     np.testing.assert_allclose(got, ref.components_, rtol=5e-2, atol=5e-3)
 
 
@@ -70,7 +70,7 @@ def test_tsvd_randomized_reconstruction():
     t = TruncatedSVD(n_components=6, algorithm="randomized").fit(
         from_numpy(X))
     Z = t.transform(from_numpy(X))
-    back = t.inverse_transform(Z).gather().numpy()
+    back = t.inverse_transform(Z).gather().cpu().numpy()
     err = np.linalg.norm(back - X) / np.linalg.norm(X)
     assert err < 0.02
 
@@ -79,19 +79,19 @@ def test_incremental_pca():
     X = _lowrank(seed=4)
     ours = IncrementalPCA(n_components=5, batch_size=300).fit(from_numpy(X))
     ref = SkPCA(n_components=5).fit(X)
-    np.testing.assert_allclose(ours.singular_values_.numpy(),
+    np.testing.assert_allclose(ours.singular_values_.cpu().numpy(),
                                ref.singular_values_, rtol=5e-2)
-    got = _align_signs(ours.components_.numpy(), ref.components_)
+    got = _align_signs(ours.components_.cpu().numpy(), ref.components_)
     np.testing.assert_allclose(np.abs(got), np.abs(ref.components_),
                                rtol=0.2, atol=0.05)
-    Z = ours.transform(from_numpy(X)).gather().numpy()
+    Z = ours.transform(from_numpy(X)).gather().cpu().numpy()
     assert Z.shape == (X.shape[0], 5)
 
 
 def _pca_world2(rank, world, X_bytes, n, d, sv_ref):
     X = np.frombuffer(X_bytes, dtype=np.float32).reshape(n, d).copy()
     ours = PCA(n_components=5, svd_solver="full").fit(from_numpy(X))
-    np.testing.assert_allclose(ours.singular_values_.numpy(), sv_ref,
+    np.testing.assert_allclose(ours.singular_values_.cpu().numpy(), sv_ref,
                                rtol=1e-3)
 
 

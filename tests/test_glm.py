@@ -32,7 +32,7 @@ def test_logreg_lbfgs_matches_sklearn():
     ours = LogisticRegression(C=1.0, solver="lbfgs", max_iter=200).fit(
         from_numpy(X), from_numpy(y))
     ref = SkLog(C=1.0, solver="lbfgs", max_iter=500).fit(X, y)
-    np.testing.assert_allclose(ours.coef_.numpy(), ref.coef_[0], rtol=0.05,
+    np.testing.assert_allclose(ours.coef_.cpu().numpy(), ref.coef_[0], rtol=0.05,
                                atol=0.02)
     np.testing.assert_allclose(ours.intercept_, ref.intercept_[0], rtol=0.05,
                                atol=0.02)
@@ -67,7 +67,7 @@ def test_linreg_matches_sklearn():
     ours = LinearRegression(C=1e8, solver="lbfgs", max_iter=500).fit(
         from_numpy(X), from_numpy(y))
     ref = SkLin().fit(X, y)
-    np.testing.assert_allclose(ours.coef_.numpy(), ref.coef_, rtol=1e-2,
+    np.testing.assert_allclose(ours.coef_.cpu().numpy(), ref.coef_, rtol=1e-2,
                                atol=1e-3)
     assert ours.score(from_numpy(X), from_numpy(y)) > 0.999
 
@@ -80,7 +80,7 @@ def test_poisson_regression_recovers():
     y = rng.poisson(lam).astype(np.float32)
     ours = PoissonRegression(C=1e6, solver="lbfgs", max_iter=300).fit(
         from_numpy(X), from_numpy(y))
-    np.testing.assert_allclose(ours.coef_.numpy(), w, atol=0.05)
+    np.testing.assert_allclose(ours.coef_.cpu().numpy(), w, atol=0.05)
     np.testing.assert_allclose(ours.intercept_, 0.5, atol=0.05)
 
 
@@ -89,7 +89,7 @@ def _logreg_world2(rank, world, X_bytes, y_bytes, n, f, coef_ref):
     y = np.frombuffer(y_bytes, dtype=np.float32).copy()
     ours = LogisticRegression(C=1.0, solver="lbfgs", max_iter=200).fit(
         from_numpy(X), from_numpy(y))
-    np.testing.assert_allclose(ours.coef_.numpy(), coef_ref, rtol=0.05,
+    np.testing.assert_allclose(ours.coef_.cpu().numpy(), coef_ref, rtol=0.05,
                                atol=0.03)
 
 

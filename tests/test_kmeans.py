@@ -23,7 +23,7 @@ def test_kmeans_matches_sklearn_quality():
     ref = SkKMeans(n_clusters=4, n_init=10, random_state=0).fit(X)
     assert ours.inertia_ <= ref.inertia_ * 1.05
     assert ours.n_iter_ >= 1
-    labels = ours.labels_.gather().numpy()
+    labels = ours.labels_.gather().cpu().numpy()
     assert labels.shape == (X.shape[0],)
     assert len(np.unique(labels)) == 4
 
@@ -40,9 +40,9 @@ def test_kmeans_explicit_init_predict_transform():
     X, _ = _blobs(seed=2)
     init = X[:4].copy()
     ours = KMeans(n_clusters=4, init=init, max_iter=50).fit(from_numpy(X))
-    pred = ours.predict(from_numpy(X)).gather().numpy()
-    np.testing.assert_array_equal(pred, ours.labels_.gather().numpy())
-    d = ours.transform(from_numpy(X)).gather().numpy()
+    pred = ours.predict(from_numpy(X)).gather().cpu().numpy()
+    np.testing.assert_array_equal(pred, ours.labels_.gather().cpu().numpy())
+    d = ours.transform(from_numpy(X)).gather().cpu().numpy()
     assert d.shape == (X.shape[0], 4)
     np.testing.assert_array_equal(d.argmin(1), pred)
     # score = -inertia at the fitted centers
@@ -53,8 +53,8 @@ def test_kmeans_deterministic():
     X, _ = _blobs(seed=3)
     a = KMeans(n_clusters=4, random_state=42).fit(from_numpy(X))
     b = KMeans(n_clusters=4, random_state=42).fit(from_numpy(X))
-    np.testing.assert_array_equal(a.cluster_centers_.numpy(),
-                                  b.cluster_centers_.numpy())
+    np.testing.assert_array_equal(a.cluster_centers_.cpu().numpy(),
+                                  b.cluster_centers_.cpu().numpy())
 
 
 def test_kmeans_joblib_roundtrip(tmp_path):
@@ -63,10 +63,10 @@ def test_kmeans_joblib_roundtrip(tmp_path):
     p = tmp_path / "km.joblib"
     joblib.dump(km, p)
     km2 = joblib.load(p)
-    np.testing.assert_array_equal(km.cluster_centers_.numpy(),
-                                  km2.cluster_centers_.numpy())
-    pred = km2.predict(from_numpy(X)).gather().numpy()
-    np.testing.assert_array_equal(pred, km.labels_.gather().numpy())
+    np.testing.assert_array_equal(km.cluster_centers_.cpu().numpy(),
+                                  km2.cluster_centers_.cpu().numpy())
+    pred = km2.predict(from_numpy(X)).gather().cpu().numpy()
+    np.testing.assert_array_equal(pred, km.labels_.gather().cpu().numpy())
 
 
 def _kmeans_world2(rank, world, X_bytes, n, f, ref_inertia):

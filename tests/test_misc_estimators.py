@@ -28,7 +28,7 @@ def test_blockwise_voting_classifier():
         assert len(clf.estimators_) == 4
         acc = clf.score(from_numpy(X), from_numpy(y))
         assert acc > 0.8
-        proba = clf.predict_proba(from_numpy(X)).gather().numpy()
+        proba = clf.predict_proba(from_numpy(X)).gather().cpu().numpy()
         assert proba.shape == (2000, 2)
 
 
@@ -79,7 +79,7 @@ def test_spectral_clustering():
     lab = rng.integers(0, k, 900)
     X = (centers[lab] + rng.standard_normal((900, 2))).astype(np.float32)
     sc = SpectralClustering(n_clusters=k, n_components=60, random_state=0)
-    got = sc.fit_predict(from_numpy(X)).gather().numpy()
+    got = sc.fit_predict(from_numpy(X)).gather().cpu().numpy()
     # clusters should align with true blobs up to permutation
     from itertools import permutations
     best = max(
@@ -93,7 +93,7 @@ def test_utils_svd_flip():
     u = torch.randn(10, 3)
     v = torch.randn(3, 5)
     u2, v2 = svd_flip(u, v)
-    np.testing.assert_allclose((u2 @ v2).numpy(), (u @ v).numpy(),
+    np.testing.assert_allclose((u2 @ v2).cpu().numpy(), (u @ v).cpu().numpy(),
                                rtol=1e-5, atol=1e-5)
 
 

@@ -54,7 +54,7 @@ def test_grid_search_logreg():
     assert gs.best_params_["C"] in (0.01, 1.0, 100.0)
     assert gs.best_score_ > 0.7
     assert len(gs.cv_results_["mean_test_score"]) == 3
-    pred = gs.predict(from_numpy(X)).gather().numpy()
+    pred = gs.predict(from_numpy(X)).gather().cpu().numpy()
     assert pred.shape == (X.shape[0],)
 
 
@@ -91,11 +91,11 @@ def test_parallel_post_fit():
     ppf = ParallelPostFit(SGDClassifier(loss="log_loss", max_iter=20,
                                         random_state=0))
     ppf.fit(from_numpy(X), from_numpy(y))
-    pred = ppf.predict(from_numpy(X)).gather().numpy()
+    pred = ppf.predict(from_numpy(X)).gather().cpu().numpy()
     assert pred.shape == (1000,)
     acc = ppf.score(from_numpy(X), from_numpy(y))
     assert acc > 0.6
-    proba = ppf.predict_proba(from_numpy(X)).gather().numpy()
+    proba = ppf.predict_proba(from_numpy(X)).gather().cpu().numpy()
     assert proba.shape == (1000, 2)
 
 
@@ -117,7 +117,7 @@ def test_incremental_search():
     search.fit(from_numpy(X), from_numpy(y))
     assert search.best_score_ > 0.6
     assert search.metadata_["n_models"] == 5
-    pred = search.predict(from_numpy(X)).gather().numpy()
+    pred = search.predict(from_numpy(X)).gather().cpu().numpy()
     assert pred.shape == (2000,)
 
 

@@ -18,12 +18,12 @@ def test_standard_scaler_matches_sklearn():
     a = _data()
     ours = StandardScaler().fit(from_numpy(a))
     ref = SkStandard().fit(a)
-    np.testing.assert_allclose(ours.mean_.numpy(), ref.mean_, rtol=1e-6,
+    np.testing.assert_allclose(ours.mean_.cpu().numpy(), ref.mean_, rtol=1e-6,
                                atol=1e-6)
-    np.testing.assert_allclose(ours.scale_.numpy(), ref.scale_, rtol=1e-5)
-    got = ours.transform(from_numpy(a)).gather().numpy()
+    np.testing.assert_allclose(ours.scale_.cpu().numpy(), ref.scale_, rtol=1e-5)
+    got = ours.transform(from_numpy(a)).gather().cpu().numpy()
     np.testing.assert_allclose(got, ref.transform(a), rtol=1e-4, atol=1e-4)
-    back = ours.inverse_transform(from_numpy(got)).gather().numpy()
+    back = ours.inverse_transform(from_numpy(got)).gather().cpu().numpy()
     np.testing.assert_allclose(back, a, rtol=1e-4, atol=1e-4)
 
 
@@ -31,7 +31,7 @@ def test_standard_scaler_zero_variance_column():
     a = _data()
     a[:, 2] = 7.0
     ours = StandardScaler().fit(from_numpy(a))
-    got = ours.transform(from_numpy(a)).gather().numpy()
+    got = ours.transform(from_numpy(a)).gather().cpu().numpy()
     assert np.allclose(got[:, 2], 0.0, atol=1e-5)
 
 
@@ -39,11 +39,11 @@ def test_minmax_scaler_matches_sklearn():
     a = _data(seed=1)
     ours = MinMaxScaler().fit(from_numpy(a))
     ref = SkMinMax().fit(a)
-    np.testing.assert_allclose(ours.data_min_.numpy(), ref.data_min_,
+    np.testing.assert_allclose(ours.data_min_.cpu().numpy(), ref.data_min_,
                                rtol=1e-6)
-    np.testing.assert_allclose(ours.data_max_.numpy(), ref.data_max_,
+    np.testing.assert_allclose(ours.data_max_.cpu().numpy(), ref.data_max_,
                                rtol=1e-6)
-    got = ours.fit_transform(from_numpy(a)).gather().numpy()
+    got = ours.fit_transform(from_numpy(a)).gather().cpu().numpy()
     np.testing.assert_allclose(got, ref.transform(a), rtol=1e-4, atol=1e-5)
 
 
@@ -51,9 +51,9 @@ def _scaler_world2(rank, world, a_bytes, shape):
     a = np.frombuffer(a_bytes, dtype=np.float32).reshape(shape).copy()
     ref = SkStandard().fit(a)
     ours = StandardScaler().fit(from_numpy(a))
-    np.testing.assert_allclose(ours.mean_.numpy(), ref.mean_, rtol=1e-6,
+    np.testing.assert_allclose(ours.mean_.cpu().numpy(), ref.mean_, rtol=1e-6,
                                atol=1e-6)
-    got = ours.transform(from_numpy(a)).gather().numpy()
+    got = ours.transform(from_numpy(a)).gather().cpu().numpy()
     np.testing.assert_allclose(got, ref.transform(a), rtol=1e-4, atol=1e-4)
 
 

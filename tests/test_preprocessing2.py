@@ -30,24 +30,24 @@ def test_robust_scaler():
     a = _data()
     ours = RobustScaler().fit(from_numpy(a))
     ref = SkRobust().fit(a)
-    np.testing.assert_allclose(ours.center_.numpy(), ref.center_, rtol=0.05,
+    np.testing.assert_allclose(ours.center_.cpu().numpy(), ref.center_, rtol=0.05,
                                atol=0.05)
-    np.testing.assert_allclose(ours.scale_.numpy(), ref.scale_, rtol=0.05)
-    got = ours.transform(from_numpy(a)).gather().numpy()
+    np.testing.assert_allclose(ours.scale_.cpu().numpy(), ref.scale_, rtol=0.05)
+    got = ours.transform(from_numpy(a)).gather().cpu().numpy()
     np.testing.assert_allclose(got, ref.transform(a), rtol=0.1, atol=0.1)
-    back = ours.inverse_transform(from_numpy(got)).gather().numpy()
+    back = ours.inverse_transform(from_numpy(got)).gather().cpu().numpy()
     np.testing.assert_allclose(back, a, rtol=1e-3, atol=1e-3)
 
 
 def test_quantile_transformer_uniform():
     a = _data(seed=1)
     qt = QuantileTransformer(n_quantiles=200).fit(from_numpy(a))
-    got = qt.transform(from_numpy(a)).gather().numpy()
+    got = qt.transform(from_numpy(a)).gather().cpu().numpy()
     assert got.min() >= 0 and got.max() <= 1
     # transformed columns should be ~uniform: mean ~0.5, var ~1/12
     assert np.allclose(got.mean(axis=0), 0.5, atol=0.02)
     assert np.allclose(got.var(axis=0), 1 / 12, atol=0.01)
-    back = qt.inverse_transform(from_numpy(got)).gather().numpy()
+    back = qt.inverse_transform(from_numpy(got)).gather().cpu().numpy()
     np.testing.assert_allclose(back, a, rtol=0.05, atol=0.05)
 
 
@@ -55,7 +55,7 @@ def test_quantile_transformer_normal():
     a = _data(seed=2)
     qt = QuantileTransformer(n_quantiles=200,
                              output_distribution="normal").fit(from_numpy(a))
-    got = qt.transform(from_numpy(a)).gather().numpy()
+    got = qt.transform(from_numpy(a)).gather().cpu().numpy()
     assert np.allclose(got.mean(axis=0), 0.0, atol=0.05)
     assert np.allclose(got.std(axis=0), 1.0, atol=0.1)
 
@@ -64,10 +64,10 @@ def test_label_encoder():
     rng = np.random.default_rng(3)
     y = rng.choice([3.0, 7.0, 11.0], size=1000).astype(np.float32)
     le = LabelEncoder().fit(from_numpy(y))
-    np.testing.assert_array_equal(le.classes_.numpy(), [3.0, 7.0, 11.0])
-    t = le.transform(from_numpy(y)).gather().numpy()
+    np.testing.assert_array_equal(le.classes_.cpu().numpy(), [3.0, 7.0, 11.0])
+    t = le.transform(from_numpy(y)).gather().cpu().numpy()
     assert set(np.unique(t)) == {0, 1, 2}
-    back = le.inverse_transform(from_numpy(t.astype(np.int64))).gather().numpy()
+    back = le.inverse_transform(from_numpy(t.astype(np.int64))).gather().cpu().numpy()
     np.testing.assert_array_equal(back, y)
 
 
@@ -76,11 +76,11 @@ def test_ordinal_and_onehot():
     X = rng.choice([0.0, 2.0, 5.0], size=(500, 3)).astype(np.float32)
     oe = OrdinalEncoder().fit(from_numpy(X))
     ref = SkOrdinal().fit(X)
-    got = oe.transform(from_numpy(X)).gather().numpy()
+    got = oe.transform(from_numpy(X)).gather().cpu().numpy()
     np.testing.assert_array_equal(got, ref.transform(X))
     oh = OneHotEncoder().fit(from_numpy(X))
     refh = SkOneHot(sparse_output=False).fit(X)
-    goth = oh.transform(from_numpy(X)).gather().numpy()
+    goth = oh.transform(from_numpy(X)).gather().cpu().numpy()
     np.testing.assert_array_equal(goth, refh.transform(X))
 
 
@@ -88,7 +88,7 @@ def test_polynomial_features():
     a = _data(n=200, f=3, seed=5)
     pf = PolynomialFeatures(degree=2).fit(from_numpy(a))
     ref = SkPoly(degree=2).fit(a)
-    got = pf.transform(from_numpy(a)).gather().numpy()
+    got = pf.transform(from_numpy(a)).gather().cpu().numpy()
     np.testing.assert_allclose(got, ref.transform(a), rtol=1e-4, atol=1e-4)
     assert list(pf.get_feature_names_out()) == list(
         ref.get_feature_names_out(["x0", "x1", "x2"]))
@@ -97,13 +97,13 @@ def test_polynomial_features():
 def test_block_transformer():
     a = _data(n=100, f=2, seed=6)
     bt = BlockTransformer(lambda t: t * 2.0)
-    got = bt.fit_transform(from_numpy(a)).gather().numpy()
+    got = bt.fit_transform(from_numpy(a)).gather().cpu().numpy()
     np.testing.assert_allclose(got, a * 2.0)
 
 
 def test_normalizer():
     a = _data(n=100, f=4, seed=7)
-    nm = Normalizer().fit_transform(from_numpy(a)).gather().numpy()
+    nm = Normalizer().fit_transform(from_numpy(a)).gather().cpu().numpy()
     np.testing.assert_allclose(np.linalg.norm(nm, axis=1), 1.0, rtol=1e-5)
 
 
@@ -116,9 +116,9 @@ def test_simple_imputer(strategy):
     kw = {"fill_value": -1.0} if strategy == "constant" else {}
     ours = SimpleImputer(strategy=strategy, **kw).fit(from_numpy(a))
     ref = SkImputer(strategy=strategy, **kw).fit(a)
-    np.testing.assert_allclose(ours.statistics_.numpy(), ref.statistics_,
+    np.testing.assert_allclose(ours.statistics_.cpu().numpy(), ref.statistics_,
                                rtol=0.06, atol=0.06)
-    got = ours.transform(from_numpy(a)).gather().numpy()
+    got = ours.transform(from_numpy(a)).gather().cpu().numpy()
     assert not np.isnan(got).any()
 
 
@@ -130,10 +130,10 @@ def test_gaussian_nb():
     X = (means[y] + rng.standard_normal((n, 2))).astype(np.float32)
     ours = GaussianNB().fit(from_numpy(X), from_numpy(y.astype(np.float32)))
     ref = SkGNB().fit(X, y)
-    np.testing.assert_allclose(ours.theta_.numpy(), ref.theta_, rtol=1e-3,
+    np.testing.assert_allclose(ours.theta_.cpu().numpy(), ref.theta_, rtol=1e-3,
                                atol=1e-3)
-    np.testing.assert_allclose(ours.var_.numpy(), ref.var_, rtol=1e-2)
+    np.testing.assert_allclose(ours.var_.cpu().numpy(), ref.var_, rtol=1e-2)
     acc = ours.score(from_numpy(X), from_numpy(y.astype(np.float32)))
     assert abs(acc - ref.score(X, y)) < 0.01
-    proba = ours.predict_proba(from_numpy(X)).gather().numpy()
+    proba = ours.predict_proba(from_numpy(X)).gather().cpu().numpy()
     np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-5)

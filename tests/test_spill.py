@@ -38,7 +38,7 @@ def test_spilled_logreg_matches_resident():
     S = spill_rows(torch.from_numpy(a), block_rows=800)
     Sy = from_numpy(y)
     lr2 = LogisticRegression(C=1.0, max_iter=100).fit(S, Sy)
-    np.testing.assert_allclose(lr1.coef_.numpy(), lr2.coef_.numpy(),
+    np.testing.assert_allclose(lr1.coef_.cpu().numpy(), lr2.coef_.cpu().numpy(),
                                rtol=1e-4, atol=1e-5)
 
 
@@ -51,7 +51,7 @@ def test_spilled_kmeans():
     S = spill_rows(torch.from_numpy(a), block_rows=640)
     km2 = KMeans(n_clusters=4, random_state=0).fit(S)
     # same seeded init + same data: same result
-    np.testing.assert_allclose(km1.cluster_centers_.numpy(),
-                               km2.cluster_centers_.numpy(), rtol=1e-4,
+    np.testing.assert_allclose(km1.cluster_centers_.cpu().numpy(),
+                               km2.cluster_centers_.cpu().numpy(), rtol=1e-4,
                                atol=1e-4)
     assert abs(km1.inertia_ - km2.inertia_) / km1.inertia_ < 1e-4

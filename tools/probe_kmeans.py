@@ -46,3 +46,11 @@ for ph, name in [(0, "loop shell only"), (1, "stage"), (2, "compute"),
                  (3, "stage+compute"), (5, "stage+update"), (7, "all")]:
     timeit(f"phases={ph} ({name})",
            lambda ph=ph: ext.kmeans_phase_probe(x, c, ph))
+
+print("--- glm phase ablation (bit0 stage, bit1 z, bit2 grad) ---")
+yv = x[:, 0].contiguous()
+wv = c[0].contiguous()
+for ph, name in [(0, "shell"), (1, "stage"), (3, "stage+z"),
+                 (5, "stage+grad"), (7, "all")]:
+    timeit(f"glm phases={ph} ({name})",
+           lambda ph=ph: ext.glm_phase_probe(x, yv, wv, ph))
