@@ -66,12 +66,19 @@ def check_is_fitted(est, attr: str) -> None:
 def _to_host(v):
     if isinstance(v, torch.Tensor):
         return ("__mi355ml_tensor__", v.detach().cpu().numpy())
+    if isinstance(v, DistArray):
+        # persist the local shard + layout; restored against the current comm
+        return ("__mi355ml_distarray__", v.local.detach().cpu().numpy(),
+                v.global_rows, v.row_offset)
     return v
 
 
 def _from_host(v, device):
-    if isinstance(v, tuple) and len(v) == 2 and v[0] == "__mi355ml_tensor__":
+    if isinstance(v, tuple) and len(v) >= 2 and v[0] == "__mi355ml_tensor__":
         return torch.from_numpy(np.ascontiguousarray(v[1])).to(device)
+    if isinstance(v, tuple) and len(v) == 4 and v[0] == "__mi355ml_distarray__":
+        local = torch.from_numpy(np.ascontiguousarray(v[1])).to(device)
+        return DistArray(local, v[2], v[3])
     return v
 
 

@@ -163,7 +163,8 @@ class DistArray:
     # ------------------------------------------------------------- linalg
     def dot(self, w: torch.Tensor) -> "DistArray":
         """X·w with replicated w — purely local per shard."""
-        return DistArray(self._as2d() @ w.to(self.local.dtype),
+        w = w.to(device=self.local.device, dtype=self.local.dtype)
+        return DistArray(self._as2d() @ w,
                          self.global_rows, self.row_offset, self.comm)
 
     def xt_dot(self, v: "DistArray", dtype: torch.dtype = torch.float64

