@@ -31,6 +31,8 @@ def _mp_entry(rank, world, port, fn, args):
     # gloo + CPU tensors even on a GPU box: multi-rank CPU tests must behave
     # identically everywhere (gloo does not support every cuda collective)
     os.environ["CUDA_VISIBLE_DEVICES"] = ""
+    os.environ["HIP_VISIBLE_DEVICES"] = ""
+    os.environ["ROCR_VISIBLE_DEVICES"] = ""
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
