@@ -198,6 +198,70 @@ class DistArray:
         """Eager framework: already resident. Kept for API parity."""
         return self
 
+    def compute(self) -> torch.Tensor:
+        """Materialize the full array (dask .compute() parity)."""
+        return self.gather()
+
+    def rebalance(self) -> "DistArray":
+        """Redistribute rows to the canonical balanced contiguous layout
+        via point-to-point sends (the C8/C9 gather_dep / rechunk analog:
+        each overlapping (src, dst) range is one send over the direct xGMI
+        link). Used after masking splits leave shards uneven."""
+        import torch.distributed as dist
+        comm = self.comm
+        if comm.world_size == 1:
+            return DistArray(self.local, self.local.shape[0], 0, comm)
+        dev = self.local.device
+        n_local = torch.tensor([self.local.shape[0]], dtype=torch.int64,
+                               device=dev)
+        counts = [int(c.item()) for c in comm.all_gather(n_local)]
+        total = sum(counts)
+        cur_start = sum(counts[: comm.rank])
+        base, rem = divmod(total, comm.world_size)
+        tgt_starts = []
+        acc = 0
+        for r in range(comm.world_size):
+            tgt_starts.append(acc)
+            acc += base + (1 if r < rem else 0)
+        tgt_starts.append(acc)
+        my_ts, my_te = tgt_starts[comm.rank], tgt_starts[comm.rank + 1]
+        tail = tuple(self.local.shape[1:])
+        out = torch.empty((my_te - my_ts,) + tail, dtype=self.dtype,
+                          device=dev)
+        ops = []
+        # sends: my current rows overlapping each dst's target range
+        cs, ce = cur_start, cur_start + self.local.shape[0]
+        for r in range(comm.world_size):
+            lo = max(cs, tgt_starts[r])
+            hi = min(ce, tgt_starts[r + 1])
+            if hi <= lo:
+                continue
+            seg = self.local[lo - cs:hi - cs].contiguous()
+            if r == comm.rank:
+                out[lo - my_ts:hi - my_ts] = seg
+            else:
+                ops.append(dist.P2POp(dist.isend, seg, r))
+        # recvs: each src's current rows overlapping my target range
+        src_start = 0
+        recv_slices = []
+        for r in range(comm.world_size):
+            s0, s1 = src_start, src_start + counts[r]
+            src_start = s1
+            lo = max(s0, my_ts)
+            hi = min(s1, my_te)
+            if hi <= lo or r == comm.rank:
+                continue
+            buf = torch.empty((hi - lo,) + tail, dtype=self.dtype,
+                              device=dev)
+            ops.append(dist.P2POp(dist.irecv, buf, r))
+            recv_slices.append((lo - my_ts, hi - my_ts, buf))
+        if ops:
+            for req in dist.batch_isend_irecv(ops):
+                req.wait()
+        for lo, hi, buf in recv_slices:
+            out[lo:hi] = buf
+        return DistArray(out, total, my_ts, comm)
+
 
 # ------------------------------------------------------------------ factories
 def shard_rows(full: torch.Tensor, comm: Optional[Communicator] = None

@@ -73,3 +73,25 @@ def _dist_moments_body(rank, world, a_bytes, shape):
 def test_dist_moments_world2(mp_runner):
     a = _data(501, 5, seed=3)  # uneven split
     mp_runner(_dist_moments_body, world=2, args=(a.tobytes(), a.shape))
+
+
+def _rebalance_body(rank, world, a_bytes, shape):
+    import torch
+    from mi355ml.core.array import DistArray, from_numpy
+    a = np.frombuffer(a_bytes, dtype=np.float32).reshape(shape).copy()
+    X = from_numpy(a)
+    # make shards uneven: rank 0 keeps 10 rows, rank 1 the rest
+    keep = 10 if rank == 0 else X.local.shape[0]
+    from mi355ml.core.array import from_local
+    un = from_local(X.local[:keep].contiguous())
+    rb = un.rebalance()
+    w = un.comm.world_size
+    assert abs(rb.local.shape[0] - un.global_rows / w) <= 1
+    # content preserved in global order
+    expect = np.concatenate([a[:10], a[shape[0] // 2:]])
+    np.testing.assert_array_equal(rb.gather().cpu().numpy(), expect)
+
+
+def test_rebalance_world2(mp_runner):
+    a = _data(400, 3, seed=9)
+    mp_runner(_rebalance_body, world=2, args=(a.tobytes(), a.shape))

@@ -107,3 +107,29 @@ def test_profiler_counters():
     c.inc("samples", 5)
     assert c.snapshot()["samples"] == 5
     device_memory()  # no GPU here: just must not raise
+
+
+def test_pairwise_metrics_vs_sklearn():
+    from sklearn.metrics.pairwise import (cosine_distances,
+                                          manhattan_distances, rbf_kernel as
+                                          sk_rbf)
+    from mi355ml.metrics import (pairwise_distances, rbf_kernel,
+                                 linear_kernel, polynomial_kernel)
+    rng = np.random.default_rng(5)
+    X = rng.standard_normal((300, 6)).astype(np.float32)
+    Y = rng.standard_normal((20, 6)).astype(np.float32)
+    Yt = torch.from_numpy(Y)
+    got = pairwise_distances(from_numpy(X), Yt, "cosine").gather().cpu().numpy()
+    np.testing.assert_allclose(got, cosine_distances(X, Y), rtol=1e-4,
+                               atol=1e-5)
+    got = pairwise_distances(from_numpy(X), Yt, "manhattan").gather().cpu().numpy()
+    np.testing.assert_allclose(got, manhattan_distances(X, Y), rtol=1e-4,
+                               atol=1e-4)
+    got = rbf_kernel(from_numpy(X), Yt, gamma=0.3).gather().cpu().numpy()
+    np.testing.assert_allclose(got, sk_rbf(X, Y, gamma=0.3), rtol=1e-4,
+                               atol=1e-5)
+    got = linear_kernel(from_numpy(X), Yt).gather().cpu().numpy()
+    np.testing.assert_allclose(got, X @ Y.T, rtol=1e-4, atol=1e-4)
+    got = polynomial_kernel(from_numpy(X), Yt, degree=2).gather().cpu().numpy()
+    ref = (X @ Y.T / 6 + 1.0) ** 2
+    np.testing.assert_allclose(got, ref, rtol=1e-3, atol=1e-4)
