@@ -133,20 +133,26 @@ __device__ __forceinline__ void vm_drain() {
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 }
 
-// LDS-DMA one KM_TR x d4 tile (KB-exact when d%4==0): wave w copies chunks
-// w, w+4, ... of 1 KiB each (64 lanes x 16 B), dest lane-linear.
-__device__ __forceinline__ void glds_tile(const float* __restrict__ src,
-                                          float* dst, int nfloats, int wave,
-                                          int lane) {
+// LDS-DMA one tile (KB-exact when d%4==0): wave w copies chunks
+// w, w+nwaves, ... of 1 KiB each (64 lanes x 16 B), dest lane-linear.
+__device__ __forceinline__ void glds_tile_w(const float* __restrict__ src,
+                                            float* dst, int nfloats,
+                                            int wave, int lane, int nwaves) {
   auto gsrc = (const __attribute__((address_space(1))) char*)(const void*)src;
   auto ldst = (__attribute__((address_space(3))) char*)(void*)dst;
   const int nchunks = nfloats >> 8;  // 256 floats = 1 KiB per chunk
-  for (int ch = wave; ch < nchunks; ch += 4) {
+  for (int ch = wave; ch < nchunks; ch += nwaves) {
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)(gsrc + ch * 1024 +
                                                         lane * 16),
         (__attribute__((address_space(3))) void*)(ldst + ch * 1024), 16, 0, 0);
   }
+}
+
+__device__ __forceinline__ void glds_tile(const float* __restrict__ src,
+                                          float* dst, int nfloats, int wave,
+                                          int lane) {
+  glds_tile_w(src, dst, nfloats, wave, lane, 4);
 }
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -355,16 +361,23 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
 // ---------------------------------------------------------------------------
 #define KM_TR 64   // rows per tile (4 waves x 16 rows)
 
-// NTILES = exact compile-time center-tile count (k16/16): keeps the
-// accumulator array statically indexed AND the MFMA inner loop branch-free —
-// runtime-guarded unrolls made hipcc round-trip every accumulator through one
-// a[0:3] AGPR window (accvgpr_write/read + s_nop 9 per MFMA, fully serial).
+// NTILES = exact compile-time center-tile count (k16/16). 512-thread blocks
+// put TWO waves on each SIMD: the pair (w, w+4) shares one 16-row group of
+// the tile and SPLITS the center tiles (T0 = ceil(NTILES/2) for the first
+// half), so one wave's fragment-read/wait bubbles hide under its partner's
+// MFMA issue (MI355X_MICROARCH "Two waves per SIMD"). Accumulator arrays
+// stay statically indexed and the MFMA loops branch-free (guarded unrolls
+// made hipcc round-trip accumulators through one AGPR window).
+#define KM_NT 512
+
 template <int UPDATE, int NTILES, int PHASES = 7>
-__global__ __launch_bounds__(NT) void kmeans_assign_k(
+__global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
     const float* __restrict__ x, const float* __restrict__ cen, long m,
     int d, int k, int* __restrict__ labels, float* __restrict__ dists,
     float* __restrict__ sums_slab, float* __restrict__ cnt_slab,
     double* __restrict__ inertia_slab) {
+  constexpr int T0 = (NTILES + 1) / 2;     // first half's tile count
+  constexpr int T1 = NTILES - T0;          // second half's tile count
   const int d4 = (d + 3) & ~3;        // K padded to a multiple of 4
   const int k16 = NTILES * 16;        // centers padded to a multiple of 16
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
@@ -373,29 +386,34 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
   float* xtbuf = nhc + k16;                      // 2 x-tile buffers
   float* xn = xtbuf + 2 * KM_TR * d4;            // row norms [KM_TR]
   int* rlab = (int*)(xn + KM_TR);                // row labels [KM_TR]
-  float* bins = (float*)(rlab + KM_TR);          // UPDATE: [k*d]
+  float* cbest = (float*)(rlab + KM_TR);         // per-half candidates
+  int* cbj = (int*)(cbest + 2 * KM_TR);          // [2][KM_TR]
+  float* bins = (float*)(cbj + 2 * KM_TR);       // UPDATE: [k*d]
   float* bcnt = bins + (UPDATE ? k * d : 0);     // UPDATE: [k]
   double* red = (double*)(smem_raw + ((((size_t)(k16 * d4 + k16 +
-      2 * KM_TR * d4 + 2 * KM_TR + (UPDATE ? (k * d + k) : 0)) * 4 + 15) & ~15)));
+      2 * KM_TR * d4 + 6 * KM_TR + (UPDATE ? (k * d + k) : 0)) * 4 + 15) &
+      ~15)));
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int lj = lane & 15;   // fragment column (center within tile)
-  const int lk = lane >> 4;   // fragment K index (0..3)
+  const int wave = tid >> 6;   // 0..7
+  const int half = wave >> 2;  // tile split
+  const int qw = wave & 3;     // row group within the tile
+  const int lj = lane & 15;    // fragment column (center within tile)
+  const int lk = lane >> 4;    // fragment K index (0..3)
 
   // stage centers (zero-padded) + norms
-  for (int i = tid; i < k16 * d4; i += NT) ct[i] = 0.0f;
+  for (int i = tid; i < k16 * d4; i += KM_NT) ct[i] = 0.0f;
   __syncthreads();
-  for (int i = tid; i < k * d; i += NT) {
+  for (int i = tid; i < k * d; i += KM_NT) {
     const int j = i / d;
     ct[j * d4 + (i - j * d)] = cen[i];
   }
   if (UPDATE) {
-    for (int i = tid; i < k * d; i += NT) bins[i] = 0.0f;
-    for (int i = tid; i < k; i += NT) bcnt[i] = 0.0f;
+    for (int i = tid; i < k * d; i += KM_NT) bins[i] = 0.0f;
+    for (int i = tid; i < k; i += KM_NT) bcnt[i] = 0.0f;
   }
   __syncthreads();
-  for (int j = tid; j < k16; j += NT) {
+  for (int j = tid; j < k16; j += KM_NT) {
     if (j < k) {
       float s = 0.0f;
       const float* cp = ct + j * d4;
@@ -409,10 +427,8 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
 
   double iacc = 0.0;
   const long tiles = (m + KM_TR - 1) / KM_TR;
-  // Double-buffered LDS-DMA staging: tile t+1 streams into buffer cur^1 by
-  // global_load_lds while tile t computes out of buffer cur; one vmcnt
-  // drain + raw barrier per tile (the §5 'Pipelining across barriers'
-  // pattern: __syncthreads would drain the DMA early).
+  // Double-buffered LDS-DMA staging: tile t+1 streams in by global_load_lds
+  // while tile t computes; raw barriers keep the DMA in flight.
   const bool use_glds = (d == d4) && ((PHASES & 1) != 0);
   int cur = 0;
   long tile = blockIdx.x;
@@ -424,23 +440,26 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       const long total4 = ((long)drows * d) >> 2;
       const float4* src4 = (const float4*)(x + dr0 * (long)d);
       float4* dst4 = (float4*)dst;
-      for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
+      for (long i = tid; i < total4; i += KM_NT) dst4[i] = src4[i];
     } else {
-      for (int i = tid; i < drows * d; i += NT) {
+      for (int i = tid; i < drows * d; i += KM_NT) {
         const int rr = i / d;
         dst[rr * d4 + (i - rr * d)] = x[dr0 * (long)d + i];
       }
-      for (int i = tid; i < drows * (d4 - d); i += NT) {
+      for (int i = tid; i < drows * (d4 - d); i += KM_NT) {
         const int rr = i / (d4 - d);
         dst[rr * d4 + d + (i - rr * (d4 - d))] = 0.0f;
       }
     }
   };
 
+  auto glds8 = [&](float* dst, long t) {
+    glds_tile_w(x + t * KM_TR * (long)d, dst, KM_TR * d4, wave, lane, 8);
+  };
+
   if ((PHASES & 1) && tile < tiles) {
-    // prologue: fill buffer 0 with this WG's first tile
     if (use_glds && m - tile * KM_TR >= KM_TR) {
-      glds_tile(x + tile * KM_TR * (long)d, xtbuf, KM_TR * d4, wave, lane);
+      glds8(xtbuf, tile);
       vm_drain();
     } else {
       stage_direct(xtbuf, tile);
@@ -451,98 +470,85 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
     const long r0 = tile * KM_TR;
     const int rows = (int)min((long)KM_TR, m - r0);
     float* xt = xtbuf + cur * KM_TR * d4;
-    // prefetch next tile into the other buffer (in flight all tile long)
     const long ntile = tile + gridDim.x;
     const bool pf = use_glds && ntile < tiles && (m - ntile * KM_TR) >= KM_TR;
-    if (pf)
-      glds_tile(x + ntile * KM_TR * (long)d, xtbuf + (cur ^ 1) * KM_TR * d4,
-                KM_TR * d4, wave, lane);
-    // ---- row norms: only the dists output (UPDATE=0) needs per-row ‖x‖²;
-    // the Lloyd path folds Σ‖x‖² into the update column pass instead.
+    if (pf) glds8(xtbuf + (cur ^ 1) * KM_TR * d4, ntile);
+    // ---- row norms: only the dists output (UPDATE=0) needs per-row ‖x‖²
     if (!UPDATE && (PHASES & 2)) {
-      const int r = tid >> 2;
-      const int q = tid & 3;
+      const int r = tid >> 3;
+      const int q = tid & 7;
       float s = 0.0f;
       if (r < rows) {
         const float* xp = xt + r * d4;
-        for (int j = q; j < d4; j += 4) s += xp[j] * xp[j];
+        for (int j = q; j < d4; j += 8) s += xp[j] * xp[j];
       }
-      s += __shfl_xor(s, 1, 4);
-      s += __shfl_xor(s, 2, 4);
+      s += __shfl_xor(s, 1, 8);
+      s += __shfl_xor(s, 2, 8);
+      s += __shfl_xor(s, 4, 8);
       if (r < rows && q == 0) xn[r] = s;
       lds_barrier();
     }
-    // ---- MFMA distance GEMM: this wave's 16 rows vs all center tiles ----
+    // ---- MFMA distance GEMM: wave pair shares 16 rows, splits tiles ----
     if (PHASES & 2) {
-      const int ri = wave * 16 + lj;  // A-fragment row for this lane
-      f32x4 acc[NTILES];  // statically indexed, branch-free unrolls
-#pragma unroll
-      for (int t = 0; t < NTILES; ++t) {
-        const float v = nhc[16 * t + lj];
-        acc[t] = (f32x4){v, v, v, v};
-      }
+      const int ri = qw * 16 + lj;
       const float* xrow = xt + ri * d4 + lk;
-      const float* cb = ct + lj * d4 + lk;
-      // Ping-pong 2-step software pipeline with NO register rotation: the
-      // rotation movs of the previous version forced lgkmcnt(0) (drain
-      // everything) each step; ping-pong buffers let hipcc emit counted
-      // lgkmcnt(N) so each step's fragment reads land a half-iteration
-      // before their MFMAs.
-      // BRANCHLESS body: guards inside the loop create phi joins that make
-      // the allocator ping-pong the accumulators VGPR<->AGPR (28 accvgpr
-      // moves per step); prefetch indices are clamped instead (an in-bounds
-      // read whose value goes unused in the tail costs nothing).
-      const int kl = d4 - 4;  // last valid step
-      float a0 = xrow[0];
-      float a1 = xrow[min(4, kl)];
-      float b0[NTILES], b1[NTILES];
-#pragma unroll
-      for (int t = 0; t < NTILES; ++t) {
-        b0[t] = cb[16 * t * d4];
-        b1[t] = cb[16 * t * d4 + min(4, kl)];
-      }
-      int k0 = 0;
-      for (; k0 + 8 <= d4; k0 += 8) {
-        // both steps' MFMAs back to back, then BOTH refills adjacent so the
-        // per-tile (kn0, kn1) reads merge into ds_read2_b32 (halves the
-        // LDS instruction count; the refills get 14 MFMA issues of slack)
-#pragma unroll
-        for (int t = 0; t < NTILES; ++t)
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],
-                                                        0, 0, 0);
-#pragma unroll
-        for (int t = 0; t < NTILES; ++t)
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1[t], acc[t],
-                                                        0, 0, 0);
-        const int kn0 = min(k0 + 8, kl);
-        const int kn1 = min(k0 + 12, kl);
-        a0 = xrow[kn0];
-        a1 = xrow[kn1];
-#pragma unroll
-        for (int t = 0; t < NTILES; ++t) {
-          b0[t] = cb[16 * t * d4 + kn0];
-          b1[t] = cb[16 * t * d4 + kn1];
-        }
-      }
-      if (k0 < d4) {  // d4 % 8 == 4 tail: one more step (data in buf0)
-#pragma unroll
-        for (int t = 0; t < NTILES; ++t)
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],
-                                                        0, 0, 0);
-      }
-      // epilogue: per-lane running argmax over center tiles, then reduce
-      // across the 16 lanes of the row group (rows (lane>>4)*4 + reg).
+      const int tbase = half ? T0 : 0;
+      const float* cb = ct + (16 * tbase + lj) * d4 + lk;
+      const int kl = d4 - 4;
       float best0 = -INFINITY, best1 = -INFINITY, best2 = -INFINITY,
             best3 = -INFINITY;
       int bj0 = 0, bj1 = 0, bj2 = 0, bj3 = 0;
-#pragma unroll
-      for (int t = 0; t < NTILES; ++t) {
-        const int j = 16 * t + lj;
-        if (acc[t][0] > best0) { best0 = acc[t][0]; bj0 = j; }
-        if (acc[t][1] > best1) { best1 = acc[t][1]; bj1 = j; }
-        if (acc[t][2] > best2) { best2 = acc[t][2]; bj2 = j; }
-        if (acc[t][3] > best3) { best3 = acc[t][3]; bj3 = j; }
+#define KM_MFMA_HALF(MYT)                                                   \
+  {                                                                         \
+    f32x4 acc[MYT];                                                         \
+    _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                       \
+      const float v = nhc[16 * (tbase + t) + lj];                           \
+      acc[t] = (f32x4){v, v, v, v};                                         \
+    }                                                                       \
+    float a0 = xrow[0];                                                     \
+    float a1 = xrow[min(4, kl)];                                            \
+    float b0[MYT], b1[MYT];                                                 \
+    _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                       \
+      b0[t] = cb[16 * t * d4];                                              \
+      b1[t] = cb[16 * t * d4 + min(4, kl)];                                 \
+    }                                                                       \
+    int k0 = 0;                                                             \
+    for (; k0 + 8 <= d4; k0 += 8) {                                         \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t)                       \
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],    \
+                                                      0, 0, 0);             \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t)                       \
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1[t], acc[t],    \
+                                                      0, 0, 0);             \
+      const int kn0 = min(k0 + 8, kl);                                      \
+      const int kn1 = min(k0 + 12, kl);                                     \
+      a0 = xrow[kn0];                                                       \
+      a1 = xrow[kn1];                                                       \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                     \
+        b0[t] = cb[16 * t * d4 + kn0];                                      \
+        b1[t] = cb[16 * t * d4 + kn1];                                      \
+      }                                                                     \
+    }                                                                       \
+    if (k0 < d4) {                                                          \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t)                       \
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],    \
+                                                      0, 0, 0);             \
+    }                                                                       \
+    _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                       \
+      const int j = 16 * (tbase + t) + lj;                                  \
+      if (acc[t][0] > best0) { best0 = acc[t][0]; bj0 = j; }                \
+      if (acc[t][1] > best1) { best1 = acc[t][1]; bj1 = j; }                \
+      if (acc[t][2] > best2) { best2 = acc[t][2]; bj2 = j; }                \
+      if (acc[t][3] > best3) { best3 = acc[t][3]; bj3 = j; }                \
+    }                                                                       \
+  }
+      constexpr int T1C = T1 > 0 ? T1 : 1;  // avoid zero-sized arrays
+      if (half == 0) {
+        KM_MFMA_HALF(T0)
+      } else if (T1 > 0) {
+        KM_MFMA_HALF(T1C)
       }
+#undef KM_MFMA_HALF
 #define KM_RED(BEST, BJ)                                                   \
   for (int off = 8; off > 0; off >>= 1) {                                  \
     const float ov = __shfl_xor(BEST, off, 16);                            \
@@ -554,20 +560,39 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
       KM_RED(best2, bj2)
       KM_RED(best3, bj3)
 #undef KM_RED
+      // publish this half's per-row candidates, then half 0 finalizes
       if (lj == 0) {
-        const int rbase = wave * 16 + lk * 4;
+        const int rbase = qw * 16 + lk * 4;
         const float bests[4] = {best0, best1, best2, best3};
         const int bjs[4] = {bj0, bj1, bj2, bj3};
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
           const int r = rbase + g;
+          cbest[half * KM_TR + r] = bests[g];
+          cbj[half * KM_TR + r] = bjs[g];
+        }
+      }
+      lds_barrier();
+      if (half == 0 && lj == 0) {
+        const int rbase = qw * 16 + lk * 4;
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          const int r = rbase + g;
           if (r < rows) {
-            labels[r0 + r] = bjs[g];
+            float bv = cbest[r];
+            int bj = cbj[r];
+            const float ov = cbest[KM_TR + r];
+            const int oj = cbj[KM_TR + r];
+            if (T1 > 0 && (ov > bv || (ov == bv && oj < bj))) {
+              bv = ov;
+              bj = oj;
+            }
+            labels[r0 + r] = bj;
             if (UPDATE) {
-              rlab[r] = bjs[g];
-              iacc += -2.0 * (double)bests[g];  // + Σ‖x‖² from update pass
+              rlab[r] = bj;
+              iacc += -2.0 * (double)bv;  // + Σ‖x‖² from the update pass
             } else {
-              dists[r0 + r] = fmaxf(xn[r] - 2.0f * bests[g], 0.0f);
+              dists[r0 + r] = fmaxf(xn[r] - 2.0f * bv, 0.0f);
             }
           }
         }
@@ -575,12 +600,9 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
     }
     if (UPDATE && (PHASES & 4)) {
       lds_barrier();  // rlab ready
-      // Column-partitioned accumulation: wave w owns columns
-      // [w*dq, (w+1)*dq); each column has exactly ONE owner lane, so plain
-      // LDS read-modify-write suffices (the per-element atomicAdd version
-      // serialized 64 scattered lanes per instruction — it was ~46% of the
-      // kernel's LDS-array time).
-      const int dq = (d + 3) >> 2;  // columns per wave
+      // column-partitioned accumulation: wave w owns columns
+      // [w*dq, (w+1)*dq) — exactly ONE owner lane per column, plain RMW
+      const int dq = (d + 7) >> 3;
       const int c = wave * dq + lane;
       if (lane < dq && c < d) {
         float sx2 = 0.0f;
@@ -623,13 +645,13 @@ __global__ __launch_bounds__(NT) void kmeans_assign_k(
 
   if (UPDATE) {
     __syncthreads();
-    for (int i = tid; i < k * d; i += NT)
+    for (int i = tid; i < k * d; i += KM_NT)
       sums_slab[(long)blockIdx.x * k * d + i] = bins[i];
-    for (int i = tid; i < k; i += NT)
+    for (int i = tid; i < k; i += KM_NT)
       cnt_slab[(long)blockIdx.x * k + i] = bcnt[i];
     red[tid] = iacc;
     __syncthreads();
-    for (int s = NT / 2; s > 0; s >>= 1) {
+    for (int s = KM_NT / 2; s > 0; s >>= 1) {
       if (tid < s) red[tid] += red[tid + s];
       __syncthreads();
     }
@@ -675,9 +697,9 @@ extern "C" __global__ __launch_bounds__(NT) void affine_transform_k(
 static size_t kmeans_smem_bytes_impl(int k, int d, int update) {
   const int d4 = (d + 3) & ~3;
   const int k16 = (k + 15) & ~15;
-  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)2 * 64 * d4 + 2 * 64;
+  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)2 * 64 * d4 + 6 * 64;
   if (update) f32s += (size_t)k * d + k;
-  return (((f32s * 4 + 15) & ~15)) + NT * sizeof(double);
+  return (((f32s * 4 + 15) & ~15)) + 512 * sizeof(double);
 }
 
 template <int UPDATE>
@@ -691,7 +713,7 @@ static void kmeans_dispatch(const float* x, const float* cen, long m, int d,
 #define KM_LAUNCH(NTL)                                                      \
   case NTL:                                                                 \
     hipLaunchKernelGGL((kmeans_assign_k<UPDATE, NTL>), dim3(nblocks),       \
-                       dim3(NT), smem, stream, x, cen, m, d, k, labels,     \
+                       dim3(KM_NT), smem, stream, x, cen, m, d, k, labels,  \
                        dists, sums_slab, cnt_slab, inertia_slab);           \
     break
   switch (ntiles) {
@@ -787,9 +809,9 @@ void launch_kmeans_probe(const float* x, const float* cen, long m, int d,
   if (((k + 15) & ~15) / 16 != 7) return;  // probe supports k=100..112 only
 #define KM_P(PH)                                                            \
   case PH:                                                                  \
-    hipLaunchKernelGGL((kmeans_assign_k<1, 7, PH>), dim3(nblocks), dim3(NT),\
-                       smem, stream, x, cen, m, d, k, labels, nullptr,      \
-                       sums_slab, cnt_slab, inertia_slab);                  \
+    hipLaunchKernelGGL((kmeans_assign_k<1, 7, PH>), dim3(nblocks),          \
+                       dim3(KM_NT), smem, stream, x, cen, m, d, k, labels,  \
+                       nullptr, sums_slab, cnt_slab, inertia_slab);         \
     break
   switch (phases) { KM_P(0); KM_P(1); KM_P(2); KM_P(3); KM_P(5); KM_P(7); }
 #undef KM_P
