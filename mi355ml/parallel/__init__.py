@@ -1,3 +1,3 @@
-from . import comm
+from . import comm, coordination
 
-__all__ = ["comm"]
+__all__ = ["comm", "coordination"]

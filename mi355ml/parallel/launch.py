@@ -17,6 +17,10 @@ def main(argv=None):
     ap.add_argument("--gpus", "-n", type=int, default=None,
                     help="ranks (default: visible GPU count, else 1)")
     ap.add_argument("--master-port", type=int, default=29531)
+    ap.add_argument("--max-restarts", type=int, default=0,
+                    help="elastic rank-respawn attempts on failure (the "
+                         "reference's nanny auto-restart analog; solvers "
+                         "resume from their checkpoints — SURVEY §5.3)")
     ap.add_argument("script")
     ap.add_argument("script_args", nargs=argparse.REMAINDER)
     args = ap.parse_args(argv)
@@ -35,6 +39,7 @@ def main(argv=None):
         f"--nproc-per-node={n}",
         "--master-addr=127.0.0.1",
         f"--master-port={args.master_port}",
+        f"--max-restarts={args.max_restarts}",
         args.script,
         *args.script_args,
     ])

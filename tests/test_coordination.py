@@ -1,0 +1,58 @@
+"""Coordination primitives (Event/Lock/Variable/Queue/Semaphore) —
+single-process and gloo world-2."""
+
+import numpy as np
+
+from mi355ml.parallel.coordination import (Event, Lock, Queue, Semaphore,
+                                           Variable)
+
+
+def test_local_primitives():
+    v = Variable("v1")
+    v.set({"a": 1})
+    assert v.get() == {"a": 1}
+    e = Event("e1")
+    assert not e.is_set()
+    e.set()
+    assert e.is_set()
+    e.wait(timeout=1)
+    with Lock("l1"):
+        pass
+    q = Queue("q1")
+    q.put(10)
+    q.put("x")
+    assert q.qsize() == 2
+    assert q.get(timeout=1) == 10
+    assert q.get(timeout=1) == "x"
+    s = Semaphore(2, "s1")
+    assert s.acquire(timeout=1)
+    assert s.acquire(timeout=1)
+    assert not s.acquire(timeout=0.1)
+    s.release()
+    assert s.acquire(timeout=1)
+    s.release(); s.release()
+
+
+def _dist_body(rank, world):
+    v = Variable("dv")
+    e = Event("de")
+    q = Queue("dq")
+    if rank == 0:
+        v.set(42)
+        q.put("from0")
+        e.set()
+    else:
+        e.wait(timeout=30)
+        assert v.get() == 42
+        assert q.get(timeout=30) == "from0"
+    with Lock("dl"):
+        val = Variable("counter")
+        try:
+            cur = val.get()
+        except Exception:
+            cur = 0
+        val.set(cur + 1)
+
+
+def test_dist_primitives_world2(mp_runner):
+    mp_runner(_dist_body, world=2)
