@@ -112,7 +112,7 @@ std::vector<torch::Tensor> kmeans_assign(torch::Tensor x,
   TORCH_CHECK(kmeans_smem_bytes(k, d, 1) <= 160 * 1024 - 1024,
               "k*d too large for LDS (", kmeans_smem_bytes(k, d, 1),
               " bytes)");
-  const int nb = (int)std::max(1L, std::min(512L, (m + 255) / 256));
+  const int nb = (int)std::max(1L, std::min(256L, (m + 255) / 256));
   auto f32 = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
   auto f64 = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
   auto labels = torch::empty({m}, torch::TensorOptions()
@@ -140,7 +140,7 @@ std::vector<torch::Tensor> closest_center(torch::Tensor x,
   TORCH_CHECK(k <= 256, "fused closest_center supports k <= 256");
   TORCH_CHECK(kmeans_smem_bytes(k, d, 0) <= 160 * 1024 - 1024,
               "k*d too large for LDS");
-  const int nb = (int)std::max(1L, std::min(2048L, (m + 255) / 256));
+  const int nb = (int)std::max(1L, std::min(256L, (m + 255) / 256));
   auto labels = torch::empty({m}, torch::TensorOptions()
                                       .dtype(torch::kInt32)
                                       .device(x.device()));
