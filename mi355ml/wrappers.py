@@ -21,12 +21,19 @@ from sklearn.base import clone
 
 from .base import BaseEstimator, check_array
 from .core.array import DistArray, from_local
+from .core.bag import DistList
 from .core import random as mlrand
 from .metrics import accuracy_score, r2_score
 
 
 def _to_numpy(t: torch.Tensor) -> np.ndarray:
     return t.detach().cpu().numpy()
+
+
+def _is_partitioned(X) -> bool:
+    """DistList of scipy-sparse/numpy partitions (text-pipeline outputs:
+    the reference's array-of-sparse-blocks with unknown row chunks)."""
+    return isinstance(X, DistList)
 
 
 def _blockwise_method(est, name, X: DistArray, out_dtype=torch.float32):
@@ -46,8 +53,23 @@ class ParallelPostFit(BaseEstimator):
         self.scoring = scoring
 
     def fit(self, X, y=None, **kwargs):
-        X = check_array(X, dtype=None)
         est = clone(self.estimator)
+        if _is_partitioned(X):
+            import scipy.sparse as sp
+            parts = X.gather()
+            Xc = sp.vstack(parts) if parts and sp.issparse(parts[0]) \
+                else np.concatenate(parts)
+            yc = None
+            if y is not None:
+                yg = (y.gather() if _is_partitioned(y)
+                      else check_array(y, dtype=None).gather())
+                yc = (np.concatenate(yg) if isinstance(yg, list)
+                      else _to_numpy(yg))
+            est.fit(Xc, yc, **kwargs) if yc is not None else est.fit(
+                Xc, **kwargs)
+            self.estimator_ = est
+            return self
+        X = check_array(X, dtype=None)
         Xc = _to_numpy(X.gather())
         if y is not None:
             y = check_array(y, dtype=None)
@@ -62,18 +84,30 @@ class ParallelPostFit(BaseEstimator):
         return self.estimator_.classes_
 
     def predict(self, X):
+        if _is_partitioned(X):
+            fn = getattr(self.estimator_, "predict")
+            return X.map(fn)
         X = check_array(X, dtype=None)
         return _blockwise_method(self.estimator_, "predict", X)
 
     def predict_proba(self, X):
+        if _is_partitioned(X):
+            fn = getattr(self.estimator_, "predict_proba")
+            return X.map(fn)
         X = check_array(X, dtype=None)
         return _blockwise_method(self.estimator_, "predict_proba", X)
 
     def predict_log_proba(self, X):
+        if _is_partitioned(X):
+            fn = getattr(self.estimator_, "predict_log_proba")
+            return X.map(fn)
         X = check_array(X, dtype=None)
         return _blockwise_method(self.estimator_, "predict_log_proba", X)
 
     def transform(self, X):
+        if _is_partitioned(X):
+            fn = getattr(self.estimator_, "transform")
+            return X.map(fn)
         X = check_array(X, dtype=None)
         return _blockwise_method(self.estimator_, "transform", X)
 
@@ -98,7 +132,7 @@ class Incremental(ParallelPostFit):
         self.random_state = random_state
         self.assume_equal_chunks = assume_equal_chunks
 
-    def _partial_fit_pass(self, est, X: DistArray, y, classes, epoch: int,
+    def _partial_fit_pass(self, est, X, y, classes, epoch: int,
                           **kwargs):
         comm = X.comm
         order = list(range(comm.world_size))
@@ -109,8 +143,19 @@ class Incremental(ParallelPostFit):
             rng.shuffle(order)
         for r in order:
             if comm.rank == r and est is not None:
-                xb = _to_numpy(X.local if X.ndim == 2 else X.local[:, None])
-                yb = _to_numpy(y.local) if y is not None else None
+                if _is_partitioned(X):
+                    import scipy.sparse as sp
+                    parts = X.local
+                    xb = (sp.vstack(parts) if parts and
+                          sp.issparse(parts[0]) else
+                          (np.concatenate(parts) if parts else
+                           np.zeros((0, 0))))
+                    yb = (np.concatenate(y.local) if _is_partitioned(y)
+                          else _to_numpy(y.local)) if y is not None else None
+                else:
+                    xb = _to_numpy(X.local if X.ndim == 2
+                                   else X.local[:, None])
+                    yb = _to_numpy(y.local) if y is not None else None
                 if xb.shape[0]:
                     if classes is not None and yb is not None:
                         # sklearn accepts (identical) classes on every call
@@ -125,15 +170,24 @@ class Incremental(ParallelPostFit):
         return est
 
     def fit(self, X, y=None, classes=None, epochs: int = 1, **kwargs):
-        X = check_array(X, dtype=None)
-        if y is not None:
+        if not _is_partitioned(X):
+            X = check_array(X, dtype=None)
+        if y is not None and not _is_partitioned(y):
             y = check_array(y, dtype=None)
         est = clone(self.estimator)
         if classes is None and y is not None and \
                 hasattr(est, "partial_fit") and \
                 est._estimator_type == "classifier":
-            from .ops.unique import unique
-            classes = _to_numpy(unique(y))
+            if _is_partitioned(y):
+                vals = set()
+                for part in y.comm.all_gather_object(
+                        sorted(set(np.concatenate(y.local).tolist()
+                                   if y.local else []))):
+                    vals.update(part)
+                classes = np.asarray(sorted(vals))
+            else:
+                from .ops.unique import unique
+                classes = _to_numpy(unique(y))
         for ep in range(epochs):
             est = self._partial_fit_pass(est, X, y, classes, ep, **kwargs)
         self.estimator_ = est

@@ -133,3 +133,22 @@ def test_pairwise_metrics_vs_sklearn():
     got = polynomial_kernel(from_numpy(X), Yt, degree=2).gather().cpu().numpy()
     ref = (X @ Y.T / 6 + 1.0) ** 2
     np.testing.assert_allclose(got, ref, rtol=1e-3, atol=1e-4)
+
+
+def test_text_pipeline_sparse_incremental():
+    """HashingVectorizer -> Incremental(SGDClassifier) over sparse
+    partitions — the reference's text workflow."""
+    from mi355ml.wrappers import Incremental, ParallelPostFit
+    docs = ["good great fine excellent", "bad awful terrible poor",
+            "great good nice", "terrible bad sad", "excellent fine happy",
+            "awful poor worst"] * 30
+    labels = np.array([1, 0, 1, 0, 1, 0] * 30, dtype=np.int64)
+    hv = HashingVectorizer(n_features=256)
+    Xs = hv.transform(DistList.from_global(docs))
+    ys = DistList.from_global([labels])
+    inc = Incremental(SGDClassifier(loss="log_loss", random_state=0),
+                      shuffle_blocks=False)
+    inc.fit(Xs, ys, epochs=4)
+    preds = inc.predict(Xs).gather()
+    pred = np.concatenate([np.asarray(p) for p in preds])
+    assert (pred == labels).mean() > 0.9
