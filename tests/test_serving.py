@@ -1,0 +1,32 @@
+"""Model server: FastAPI app over a fitted estimator (TestClient, no net)."""
+
+import numpy as np
+from fastapi.testclient import TestClient
+
+from mi355ml.core.array import from_numpy
+from mi355ml.linear_model import LogisticRegression
+from mi355ml.serving import create_app
+
+
+def test_serving_roundtrip():
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((500, 4)).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] > 0).astype(np.float32)
+    lr = LogisticRegression(C=10.0, max_iter=50).fit(from_numpy(X),
+                                                     from_numpy(y))
+    client = TestClient(create_app(lr))
+    assert client.get("/health").json() == {"status": "ok"}
+    info = client.get("/info").json()
+    assert info["model"] == "LogisticRegression"
+    assert info["n_features_in"] == 4
+    r = client.post("/predict", json={"data": X[:10].tolist()})
+    assert r.status_code == 200
+    pred = np.asarray(r.json()["result"])
+    assert pred.shape == (10,)
+    r = client.post("/predict_proba", json={"data": X[:5].tolist()})
+    proba = np.asarray(r.json()["result"])
+    assert proba.shape == (5, 2)
+    np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-5)
+    # malformed input -> 4xx, not a crash
+    r = client.post("/predict", json={"data": "nope"})
+    assert r.status_code == 422
