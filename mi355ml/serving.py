@@ -1,10 +1,12 @@
 """Model serving: a batch-inference HTTP server for fitted estimators.
 
 The reference has no serving story beyond joblib persistence; production
-deployment needs one. `create_app(model)` builds a FastAPI app exposing
-/health, /info, /predict and /predict_proba over a joblib-loadable
-mi355ml (or sklearn) estimator; inference runs on this process's device
-through the estimator's blockwise predict path.
+deployment needs one. `create_app(model)` builds a Starlette ASGI app
+exposing /health, /info, /predict, /predict_proba and /transform over a
+joblib-loadable mi355ml (or sklearn) estimator; inference runs on this
+process's device through the estimator's blockwise predict path.
+(Plain starlette, not FastAPI: the image's fastapi/starlette versions are
+mutually incompatible.)
 
     python -m mi355ml.serving model.joblib --host 0.0.0.0 --port 8000
 """
@@ -32,56 +34,54 @@ def _to_plain(res) -> List:
 
 
 def create_app(model: Any):
-    from fastapi import FastAPI, HTTPException
-    from pydantic import BaseModel
+    from starlette.applications import Starlette
+    from starlette.responses import JSONResponse
+    from starlette.routing import Route
 
-    class PredictRequest(BaseModel):
-        data: List[List[float]]
+    async def health(request):
+        return JSONResponse({"status": "ok"})
 
-    app = FastAPI(title="mi355ml model server")
-    app.state.model = model
+    async def info(request):
+        return JSONResponse({
+            "model": type(model).__name__,
+            "n_features_in": getattr(model, "n_features_in_", None),
+            "params": {k: repr(v) for k, v in model.get_params().items()}
+            if hasattr(model, "get_params") else {},
+        })
 
-    @app.get("/health")
-    def health():
-        return {"status": "ok"}
+    def _method(name: str):
+        async def endpoint(request):
+            if not hasattr(model, name):
+                return JSONResponse({"detail": f"model has no {name}()"},
+                                    status_code=400)
+            try:
+                payload = await request.json()
+            except Exception:
+                return JSONResponse({"detail": "invalid JSON body"},
+                                    status_code=422)
+            data = payload.get("data") if isinstance(payload, dict) else None
+            try:
+                x = np.asarray(data, dtype=np.float32)
+                assert x.ndim == 2
+            except Exception:
+                return JSONResponse({"detail": "data must be a 2-D float "
+                                               "array under key 'data'"},
+                                    status_code=422)
+            try:
+                res = getattr(model, name)(x)
+            except Exception as e:  # surface model errors as 400s
+                return JSONResponse({"detail": str(e)}, status_code=400)
+            return JSONResponse({"result": _to_plain(res)})
 
-    @app.get("/info")
-    def info():
-        m = app.state.model
-        return {
-            "model": type(m).__name__,
-            "n_features_in": getattr(m, "n_features_in_", None),
-            "params": {k: repr(v) for k, v in m.get_params().items()}
-            if hasattr(m, "get_params") else {},
-        }
+        return endpoint
 
-    def _run(name: str, req: PredictRequest):
-        m = app.state.model
-        if not hasattr(m, name):
-            raise HTTPException(status_code=400,
-                                detail=f"model has no {name}()")
-        x = np.asarray(req.data, dtype=np.float32)
-        if x.ndim != 2:
-            raise HTTPException(status_code=422, detail="data must be 2-D")
-        try:
-            res = getattr(m, name)(x)
-        except Exception as e:  # surface model errors as 400s
-            raise HTTPException(status_code=400, detail=str(e))
-        return {"result": _to_plain(res)}
-
-    @app.post("/predict")
-    def predict(req: PredictRequest):
-        return _run("predict", req)
-
-    @app.post("/predict_proba")
-    def predict_proba(req: PredictRequest):
-        return _run("predict_proba", req)
-
-    @app.post("/transform")
-    def transform(req: PredictRequest):
-        return _run("transform", req)
-
-    return app
+    return Starlette(routes=[
+        Route("/health", health, methods=["GET"]),
+        Route("/info", info, methods=["GET"]),
+        Route("/predict", _method("predict"), methods=["POST"]),
+        Route("/predict_proba", _method("predict_proba"), methods=["POST"]),
+        Route("/transform", _method("transform"), methods=["POST"]),
+    ])
 
 
 def main(argv: Optional[list] = None):

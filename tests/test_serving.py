@@ -1,7 +1,7 @@
 """Model server: FastAPI app over a fitted estimator (TestClient, no net)."""
 
 import numpy as np
-from fastapi.testclient import TestClient
+from starlette.testclient import TestClient
 
 from mi355ml.core.array import from_numpy
 from mi355ml.linear_model import LogisticRegression
