@@ -160,6 +160,59 @@ class DistArray:
             self.comm.all_reduce(mx, op="max")
         return self._vec(mx)
 
+    def histogram(self, bins: int = 10, range: Optional[Tuple[float, float]]
+                  = None) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Global histogram (dask/array/routines.py:824 parity): per-shard
+        torch.histc + ONE allreduce of the counts."""
+        flat = self.local.reshape(-1)
+        if range is None:
+            lo = float(self.min() if self.ndim == 1 else
+                       self.min().min())
+            hi = float(self.max() if self.ndim == 1 else
+                       self.max().max())
+        else:
+            lo, hi = range
+        if flat.numel():
+            h = torch.histc(flat.float(), bins=bins, min=lo, max=hi)
+        else:
+            h = torch.zeros(bins, device=self.local.device)
+        h = h.double()
+        if self.comm.world_size > 1:
+            self.comm.all_reduce(h)
+        edges = torch.linspace(lo, hi, bins + 1, dtype=torch.float64,
+                               device=self.local.device)
+        return h.long(), edges
+
+    def topk(self, k: int, largest: bool = True) -> torch.Tensor:
+        """Global top-k of a 1-D array (dask/array/reductions.py:1464):
+        per-shard topk + all-gather of the W*k candidates + final topk."""
+        assert self.ndim == 1
+        kk = min(k, self.local.shape[0])
+        vals, _ = (torch.topk(self.local, kk, largest=largest)
+                   if kk else (torch.empty(0, dtype=self.dtype,
+                                           device=self.device), None))
+        if self.comm.world_size > 1:
+            vals = torch.cat(self.comm.all_gather_varlen(vals.contiguous()))
+        out, _ = torch.topk(vals, min(k, vals.shape[0]), largest=largest)
+        return out
+
+    def __getitem__(self, idx):
+        """Global row-range slicing: X[a:b] -> DistArray over the
+        intersection of [a, b) with each rank's shard."""
+        if not isinstance(idx, slice) or idx.step not in (None, 1):
+            raise TypeError("only contiguous row slices are supported")
+        a = 0 if idx.start is None else idx.start
+        b = self.global_rows if idx.stop is None else idx.stop
+        a = max(0, a + self.global_rows if a < 0 else a)
+        b = min(self.global_rows, b + self.global_rows if b < 0 else b)
+        lo = self.row_offset
+        hi = lo + self.local.shape[0]
+        s0 = min(max(a - lo, 0), self.local.shape[0])
+        s1 = min(max(b - lo, 0), self.local.shape[0])
+        local = self.local[s0:s1]
+        return DistArray(local, max(b - a, 0),
+                         max(lo + s0 - a, 0), self.comm)
+
     # ------------------------------------------------------------- linalg
     def dot(self, w: torch.Tensor) -> "DistArray":
         """X·w with replicated w — purely local per shard."""

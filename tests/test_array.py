@@ -95,3 +95,38 @@ def _rebalance_body(rank, world, a_bytes, shape):
 def test_rebalance_world2(mp_runner):
     a = _data(400, 3, seed=9)
     mp_runner(_rebalance_body, world=2, args=(a.tobytes(), a.shape))
+
+
+def test_histogram_topk_getitem():
+    rng = np.random.default_rng(11)
+    a = rng.standard_normal(5000).astype(np.float32)
+    y = from_numpy(a)
+    h, edges = y.histogram(bins=20, range=(-4, 4))
+    ref, redges = np.histogram(a, bins=20, range=(-4, 4))
+    np.testing.assert_array_equal(h.cpu().numpy(), ref)
+    np.testing.assert_allclose(edges.cpu().numpy(), redges, rtol=1e-6)
+    tk = y.topk(7).cpu().numpy()
+    np.testing.assert_allclose(tk, np.sort(a)[-7:][::-1], rtol=1e-6)
+    tk = y.topk(3, largest=False).cpu().numpy()
+    np.testing.assert_allclose(tk, np.sort(a)[:3], rtol=1e-6)
+    M = rng.standard_normal((100, 4)).astype(np.float32)
+    X = from_numpy(M)
+    sl = X[10:37]
+    assert sl.shape == (27, 4)
+    np.testing.assert_array_equal(sl.gather().cpu().numpy(), M[10:37])
+
+
+def _hist_world2(rank, world, a_bytes, n):
+    a = np.frombuffer(a_bytes, dtype=np.float32).copy()
+    y = from_numpy(a)
+    h, _ = y.histogram(bins=10, range=(-3, 3))
+    ref, _ = np.histogram(a, bins=10, range=(-3, 3))
+    np.testing.assert_array_equal(h.cpu().numpy(), ref)
+    sl = y[100:301]
+    np.testing.assert_array_equal(sl.gather().cpu().numpy(), a[100:301])
+
+
+def test_hist_world2(mp_runner):
+    rng = np.random.default_rng(12)
+    a = rng.standard_normal(1001).astype(np.float32)
+    mp_runner(_hist_world2, world=2, args=(a.tobytes(), len(a)))
