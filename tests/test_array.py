@@ -104,7 +104,8 @@ def test_histogram_topk_getitem():
     h, edges = y.histogram(bins=20, range=(-4, 4))
     ref, redges = np.histogram(a, bins=20, range=(-4, 4))
     np.testing.assert_array_equal(h.cpu().numpy(), ref)
-    np.testing.assert_allclose(edges.cpu().numpy(), redges, rtol=1e-6)
+    np.testing.assert_allclose(edges.cpu().numpy(), redges, rtol=1e-6,
+                               atol=1e-12)
     tk = y.topk(7).cpu().numpy()
     np.testing.assert_allclose(tk, np.sort(a)[-7:][::-1], rtol=1e-6)
     tk = y.topk(3, largest=False).cpu().numpy()
