@@ -10,7 +10,7 @@ produces portable files (SURVEY §2.1 model-persistence contract).
 
 from __future__ import annotations
 
-from typing import Optional, Union
+from typing import Union
 
 import numpy as np
 import torch

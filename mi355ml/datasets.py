@@ -9,7 +9,7 @@ each rank generates its own shard").
 
 from __future__ import annotations
 
-from typing import Optional, Tuple, Union
+from typing import Tuple, Union
 
 import torch
 

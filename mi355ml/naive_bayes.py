@@ -12,7 +12,6 @@ import math
 import torch
 
 from .base import BaseEstimator, ClassifierMixin, check_X_y, check_array, check_is_fitted
-from .core.array import DistArray
 from .metrics import accuracy_score
 from .ops.unique import unique
 

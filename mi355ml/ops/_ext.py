@@ -10,8 +10,6 @@ config "kernels.require_hip" is False (CPU-only test runs).
 from __future__ import annotations
 
 import importlib
-from typing import Optional
-
 import torch
 
 from .. import config

@@ -13,14 +13,12 @@ one tiny object broadcast (same pattern as IncrementalPCA).
 
 from __future__ import annotations
 
-from typing import Optional
-
 import numpy as np
 import torch
 from sklearn.base import clone
 
 from .base import BaseEstimator, check_array
-from .core.array import DistArray, from_local
+from .core.array import DistArray
 from .core.bag import DistList
 from .core import random as mlrand
 from .metrics import accuracy_score, r2_score
