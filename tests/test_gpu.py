@@ -163,3 +163,42 @@ def test_kmeans_e2e_gpu():
     assert km.n_iter_ >= 1
     # blobs are well separated: inertia per point ~ d * std^2
     assert km.inertia_ / X.global_rows < 2.0 * 50
+
+
+def test_kmeans_kernel_shape_sweep():
+    """Exercise every NTILES band + odd d (pad paths) + big k fallback."""
+    from mi355ml.ops import kmeans_assign, closest_center
+    g = torch.Generator().manual_seed(21)
+    for m, d, k in [(5000, 3, 2), (4097, 8, 16), (6000, 33, 17),
+                    (8000, 64, 64), (6000, 128, 100), (4000, 50, 256),
+                    (3000, 20, 240)]:
+        x = torch.randn(m, d, generator=g).float()
+        c = torch.randn(k, d, generator=g).float()
+        lab_ref, sums_ref, cnt_ref, in_ref = kmeans_assign(x, c)
+        lab, sums, cnt, inertia = kmeans_assign(x.cuda(), c.cuda())
+        agree = (lab.cpu() == lab_ref).float().mean()
+        assert agree > 0.995, (m, d, k, float(agree))
+        torch.testing.assert_close(cnt.cpu().sum(), cnt_ref.sum())
+        torch.testing.assert_close(inertia.cpu(), in_ref, rtol=1e-4,
+                                   atol=1e-1)
+        torch.testing.assert_close(sums.cpu().sum(0), sums_ref.sum(0),
+                                   rtol=1e-4, atol=1e-1)
+        labc, dc = closest_center(x.cuda(), c.cuda())
+        assert (labc.cpu() == lab_ref).float().mean() > 0.995
+
+
+def test_glm_kernel_width_sweep():
+    from mi355ml.ops import glm_value_grad
+    g = torch.Generator().manual_seed(22)
+    for m, d in [(3000, 4), (4096, 32), (5000, 100), (2000, 200),
+                 (1500, 384), (1024, 512)]:
+        x = torch.randn(m, d, generator=g).float()
+        y = (torch.rand(m, generator=g) < 0.5).float()
+        w = (torch.randn(d, generator=g) * 0.05).float()
+        b = torch.tensor(0.05)
+        for fam in ("logistic", "normal"):
+            l_ref, gw_ref, gb_ref = glm_value_grad(x, y, w, b, fam)
+            l_, gw, gb = glm_value_grad(x.cuda(), y.cuda(), w.cuda(), b, fam)
+            torch.testing.assert_close(l_.cpu(), l_ref, rtol=5e-6, atol=1e-2)
+            torch.testing.assert_close(gw.cpu(), gw_ref, rtol=1e-5,
+                                       atol=1e-2)
