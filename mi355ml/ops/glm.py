@@ -36,11 +36,16 @@ def glm_value_grad(x: torch.Tensor, y: torch.Tensor, w: torch.Tensor,
     assert family in _FAMILY_ID, family
     assert x.dim() == 2 and y.dim() == 1 and w.dim() == 1
     if use_hip(x):
-        ext = hip_ext()
-        loss, gw, gb = ext.glm_value_grad(
-            x.contiguous(), y.contiguous(), w.contiguous(),
-            float(b), _FAMILY_ID[family])
-        return loss, gw, gb
+        if x.shape[1] <= 512:
+            ext = hip_ext()
+            loss, gw, gb = ext.glm_value_grad(
+                x.contiguous(), y.contiguous(), w.contiguous(),
+                float(b), _FAMILY_ID[family])
+            return loss, gw, gb
+        # wide-feature path (d > the fused kernel's LDS tile): explicit
+        # rocBLAS-composed GEMV pair (plain library GEMMs, sanctioned) with
+        # the link math in f32 on device and f64 partials.
+        return _composed_value_grad(x, y, w, float(b), family)
     xd = x.to(torch.float64)
     yd = y.to(torch.float64)
     wd = w.to(torch.float64)
@@ -58,6 +63,25 @@ def glm_value_grad(x: torch.Tensor, y: torch.Tensor, w: torch.Tensor,
         r = ez - yd
     gw = xd.t() @ r
     gb = r.sum()
+    return loss, gw, gb
+
+
+def _composed_value_grad(x, y, w, b, family):
+    z = x @ w + b
+    if family == "logistic":
+        loss = (torch.nn.functional.softplus(z).double()
+                - y.double() * z.double()).sum()
+        r = torch.sigmoid(z) - y
+    elif family == "normal":
+        diff = (z - y)
+        loss = 0.5 * (diff.double() * diff.double()).sum()
+        r = diff
+    else:  # poisson
+        ez = torch.exp(z.double())
+        loss = (ez - y.double() * z.double()).sum()
+        r = (ez - y.double()).float()
+    gw = (x.t() @ r).double()
+    gb = r.double().sum()
     return loss, gw, gb
 
 

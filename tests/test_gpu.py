@@ -202,3 +202,18 @@ def test_glm_kernel_width_sweep():
             torch.testing.assert_close(l_.cpu(), l_ref, rtol=5e-6, atol=1e-2)
             torch.testing.assert_close(gw.cpu(), gw_ref, rtol=1e-5,
                                        atol=1e-2)
+
+
+def test_glm_wide_features_gpu():
+    """d > 512 routes through the rocBLAS-composed path (documented)."""
+    from mi355ml.ops import glm_value_grad
+    g = torch.Generator().manual_seed(23)
+    m, d = 2000, 700
+    x = torch.randn(m, d, generator=g).float()
+    y = (torch.rand(m, generator=g) < 0.5).float()
+    w = (torch.randn(d, generator=g) * 0.05).float()
+    b = torch.tensor(0.1)
+    l_ref, gw_ref, gb_ref = glm_value_grad(x, y, w, b, "logistic")
+    l_, gw, gb = glm_value_grad(x.cuda(), y.cuda(), w.cuda(), b, "logistic")
+    torch.testing.assert_close(l_.cpu(), l_ref, rtol=1e-4, atol=1e-2)
+    torch.testing.assert_close(gw.cpu(), gw_ref, rtol=1e-3, atol=1e-2)
