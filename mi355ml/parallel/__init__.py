@@ -1,3 +1,3 @@
-from . import comm, coordination
+from . import comm, coordination, futures
 
-__all__ = ["comm", "coordination"]
+__all__ = ["comm", "coordination", "futures"]

@@ -56,3 +56,31 @@ def _dist_body(rank, world):
 
 def test_dist_primitives_world2(mp_runner):
     mp_runner(_dist_body, world=2)
+
+
+def test_futures_local():
+    from mi355ml.parallel.futures import FuturePool
+    pool = FuturePool("t1")
+    futs = pool.map(lambda x: x * x, range(5))
+    assert pool.gather(futs) == [0, 1, 4, 9, 16]
+    f = pool.submit(lambda: 1 / 0)
+    pool.run()
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="failed on rank"):
+        f.result()
+
+
+def _futures_body(rank, world):
+    from mi355ml.parallel.futures import FuturePool
+    pool = FuturePool("t2")
+    futs = [pool.submit(lambda x=x: (x, __import__("os").getpid()))
+            for x in range(6)]
+    pool.run()
+    vals = [f.result(timeout=30) for f in futs]
+    assert [v[0] for v in vals] == list(range(6))
+    # tasks actually ran on different processes (round-robin owners)
+    assert len({v[1] for v in vals}) == world
+
+
+def test_futures_world2(mp_runner):
+    mp_runner(_futures_body, world=2)
