@@ -370,7 +370,7 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
 // made hipcc round-trip accumulators through one AGPR window).
 #define KM_NT 512
 
-template <int UPDATE, int NTILES, int PHASES = 7>
+template <int UPDATE, int NTILES, int PHASES = 15>
 __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
     const float* __restrict__ x, const float* __restrict__ cen, long m,
     int d, int k, int* __restrict__ labels, float* __restrict__ dists,
@@ -549,6 +549,11 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
         KM_MFMA_HALF(T1C)
       }
 #undef KM_MFMA_HALF
+      if (!(PHASES & 8)) {
+        // ablation arm: keep the argmax inputs live, skip the epilogue
+        asm volatile("" :: "v"(best0), "v"(best1), "v"(best2), "v"(best3),
+                     "v"(bj0), "v"(bj1), "v"(bj2), "v"(bj3));
+      } else {
 #define KM_RED(BEST, BJ)                                                   \
   for (int off = 8; off > 0; off >>= 1) {                                  \
     const float ov = __shfl_xor(BEST, off, 16);                            \
@@ -597,6 +602,7 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
           }
         }
       }
+      }  // PHASES & 8 (epilogue)
     }
     if (UPDATE && (PHASES & 4)) {
       lds_barrier();  // rlab ready
@@ -813,7 +819,8 @@ void launch_kmeans_probe(const float* x, const float* cen, long m, int d,
                        dim3(KM_NT), smem, stream, x, cen, m, d, k, labels,  \
                        nullptr, sums_slab, cnt_slab, inertia_slab);         \
     break
-  switch (phases) { KM_P(0); KM_P(1); KM_P(2); KM_P(3); KM_P(5); KM_P(7); }
+  switch (phases) { KM_P(0); KM_P(1); KM_P(3); KM_P(11);
+                    KM_P(13); KM_P(15); }
 #undef KM_P
 }
 

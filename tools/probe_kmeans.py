@@ -42,8 +42,10 @@ timeit("glm_value_grad", lambda: ext.glm_value_grad(
     x, x[:, 0].contiguous(), c[0].contiguous(), 0.1, 0))
 
 print("--- phase ablation (bit0 stage, bit1 compute, bit2 update) ---")
-for ph, name in [(0, "loop shell only"), (1, "stage"), (2, "compute"),
-                 (3, "stage+compute"), (5, "stage+update"), (7, "all")]:
+for ph, name in [(0, "loop shell only"), (1, "stage"),
+                 (3, "stage+mfma (epilogue dead)"),
+                 (11, "stage+mfma+epilogue"),
+                 (13, "stage+epi?+update"), (15, "all")]:
     timeit(f"phases={ph} ({name})",
            lambda ph=ph: ext.kmeans_phase_probe(x, c, ph))
 
