@@ -70,7 +70,6 @@ class SpectralClustering(BaseEstimator):
         Ainv_half = (evecs * (evals ** -0.5)[None, :]) @ evecs.t()
         # embedding rows per shard: K(x, sample) @ A^{-1/2}[:, :k]
         k = self.n_clusters
-        _, top = torch.linalg.eigh(A)  # ascending
         proj = (Ainv_half @ evecs[:, -k:]).to(X.dtype)
 
         def _embed(t: torch.Tensor) -> torch.Tensor:

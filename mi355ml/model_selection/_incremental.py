@@ -58,10 +58,13 @@ class BaseIncrementalSearchCV(BaseEstimator):
         self.scores_per_fit = scores_per_fit
         self.verbose = verbose
 
-    # -- subclass hook ----------------------------------------------------
+    # -- subclass hooks ---------------------------------------------------
     def _adapt(self, trials: List[_Trial], round_idx: int) -> Dict[int, int]:
         """Return {tid: n_partial_fits_next_round} for survivors."""
         raise NotImplementedError
+
+    def _initial_budget(self) -> int:
+        return 1
 
     def _initial_candidates(self) -> List[Dict]:
         n = self.n_initial_parameters
@@ -108,9 +111,8 @@ class BaseIncrementalSearchCV(BaseEstimator):
         for t in owned.values():
             t.model = clone(self.estimator).set_params(**t.params)
 
-        budgets = {t.tid: 1 for t in trials}
+        budgets = {t.tid: self._initial_budget() for t in trials}
         round_idx = 0
-        meta_pf_calls = 0
         while budgets:
             # run this round's partial_fits + scores for my trials
             local_scores = []
@@ -138,7 +140,6 @@ class BaseIncrementalSearchCV(BaseEstimator):
                     t.score = sc
                     t.pf_calls = pf
                     t.history.append(sc)
-                    meta_pf_calls += pf
                     if sc > t.best_score + self.tol:
                         t.best_score = sc
                         t.since_best = 0
@@ -181,7 +182,8 @@ class BaseIncrementalSearchCV(BaseEstimator):
             for t in trials]
         self.n_iter_ = round_idx
         self.metadata_ = {"n_models": len(trials),
-                          "partial_fit_calls": meta_pf_calls}
+                          "partial_fit_calls":
+                              int(sum(t.pf_calls for t in trials))}
         return self
 
     def predict(self, X):

@@ -20,6 +20,9 @@ class SuccessiveHalvingSearchCV(BaseIncrementalSearchCV):
         self.aggressiveness = aggressiveness
         self.n_initial_iter = n_initial_iter
 
+    def _initial_budget(self) -> int:
+        return int(self.n_initial_iter or 1)
+
     def _budget(self, rung: int) -> int:
         r0 = self.n_initial_iter or 1
         return int(r0 * self.aggressiveness ** rung)
