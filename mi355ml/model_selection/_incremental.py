@@ -227,13 +227,11 @@ class IncrementalSearchCV(BaseIncrementalSearchCV):
                 / (round_idx + 1) ** self.decay_rate)))
             alive = alive[:keep]
         if self.patience:
-            alive = [t for t in alive if t.since_best < self.patience] or \
-                alive[:1]
+            # a plateaued trial stops training (reference patience
+            # semantics); an empty dict ends the search
+            alive = [t for t in alive if t.since_best < self.patience]
         out = {}
         for t in alive:
             if t.pf_calls < self.max_iter:
                 out[t.tid] = self.fits_per_score
-        if len(alive) <= 1 and all(t.pf_calls >= self.max_iter
-                                   for t in alive):
-            return {}
         return out
