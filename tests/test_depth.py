@@ -138,3 +138,22 @@ def test_config_layering(monkeypatch):
     with pytest.raises(KeyError):
         mi355ml.config.get("no.such.key")
     assert mi355ml.config.get("no.such.key", 7) == 7
+
+
+def test_column_transformer():
+    from mi355ml.compose import ColumnTransformer, make_column_transformer
+    from mi355ml.preprocessing import MinMaxScaler, StandardScaler
+    rng = np.random.default_rng(5)
+    X = (rng.standard_normal((400, 5)) * [1, 10, 100, 5, 2]).astype(
+        np.float32)
+    ct = ColumnTransformer(
+        [("a", StandardScaler(), [0, 1]), ("b", MinMaxScaler(), [2])],
+        remainder="passthrough")
+    out = ct.fit_transform(from_numpy(X)).gather().cpu().numpy()
+    assert out.shape == (400, 5)
+    np.testing.assert_allclose(out[:, :2].mean(axis=0), 0, atol=1e-4)
+    assert out[:, 2].min() >= 0 and out[:, 2].max() <= 1
+    np.testing.assert_allclose(out[:, 3:], X[:, 3:], rtol=1e-6)
+    ct2 = make_column_transformer((StandardScaler(), [0]), remainder="drop")
+    out2 = ct2.fit_transform(from_numpy(X)).gather().cpu().numpy()
+    assert out2.shape == (400, 1)
