@@ -152,7 +152,7 @@ def test_column_transformer():
     out = ct.fit_transform(from_numpy(X)).gather().cpu().numpy()
     assert out.shape == (400, 5)
     np.testing.assert_allclose(out[:, :2].mean(axis=0), 0, atol=1e-4)
-    assert out[:, 2].min() >= 0 and out[:, 2].max() <= 1
+    assert out[:, 2].min() >= -1e-6 and out[:, 2].max() <= 1 + 1e-6
     np.testing.assert_allclose(out[:, 3:], X[:, 3:], rtol=1e-6)
     ct2 = make_column_transformer((StandardScaler(), [0]), remainder="drop")
     out2 = ct2.fit_transform(from_numpy(X)).gather().cpu().numpy()
