@@ -12,8 +12,7 @@ void launch_col_moments(const float*, long, int, double*, double*, double*,
 void launch_col_minmax(const float*, long, int, float*, float*, int,
                        hipStream_t);
 void launch_glm_value_grad(const float*, const float*, const float*, float,
-                           long, int, int, double*, double*, double*, int,
-                           hipStream_t);
+                           long, int, int, double*, int, hipStream_t);
 void launch_kmeans_assign(const float*, const float*, long, int, int, int*,
                           float*, float*, double*, int, hipStream_t);
 void launch_closest_center(const float*, const float*, long, int, int, int*,
@@ -23,7 +22,7 @@ void launch_affine_transform(const float*, const float*, const float*,
 void launch_kmeans_probe(const float*, const float*, long, int, int, int*,
                          float*, float*, double*, int, int, hipStream_t);
 void launch_glm_probe(const float*, const float*, const float*, float, long,
-                      int, int, double*, double*, double*, int, hipStream_t);
+                      int, int, double*, int, hipStream_t);
 size_t kmeans_smem_bytes(int k, int d, int update);
 }
 
@@ -87,15 +86,14 @@ std::vector<torch::Tensor> glm_value_grad(torch::Tensor x, torch::Tensor y,
   const long tiles = (m + 63) / 64;
   const int nb = (int)std::max(1L, std::min((long)kMaxBlocks, tiles));
   auto opts = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
-  auto loss_slab = torch::zeros({nb}, opts);
-  auto gw_slab = torch::zeros({nb, d}, opts);
-  auto gb_slab = torch::zeros({nb}, opts);
+  // one fused slab [nb, d+2] = (gw | loss | gb): ONE reduce launch
+  auto slab = torch::zeros({nb, d + 2}, opts);
   launch_glm_value_grad(x.data_ptr<float>(), y.data_ptr<float>(),
                         w.data_ptr<float>(), (float)b, m, d, (int)family,
-                        loss_slab.data_ptr<double>(),
-                        gw_slab.data_ptr<double>(),
-                        gb_slab.data_ptr<double>(), nb, cur_stream());
-  return {loss_slab.sum(), gw_slab.sum(0), gb_slab.sum()};
+                        slab.data_ptr<double>(), nb, cur_stream());
+  auto total = slab.sum(0);
+  using torch::indexing::Slice;
+  return {total.index({d}), total.index({Slice(0, d)}), total.index({d + 1})};
 }
 
 std::vector<torch::Tensor> kmeans_assign(torch::Tensor x,
@@ -160,14 +158,11 @@ torch::Tensor glm_phase_probe(torch::Tensor x, torch::Tensor y,
   const long tiles = (m + 63) / 64;
   const int nb = (int)std::max(1L, std::min((long)kMaxBlocks, tiles));
   auto opts = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
-  auto loss_slab = torch::zeros({nb}, opts);
-  auto gw_slab = torch::zeros({nb, d}, opts);
-  auto gb_slab = torch::zeros({nb}, opts);
+  auto slab = torch::zeros({nb, d + 2}, opts);
   launch_glm_probe(x.data_ptr<float>(), y.data_ptr<float>(),
                    w.data_ptr<float>(), 0.1f, m, d, (int)phases,
-                   loss_slab.data_ptr<double>(), gw_slab.data_ptr<double>(),
-                   gb_slab.data_ptr<double>(), nb, cur_stream());
-  return loss_slab;
+                   slab.data_ptr<double>(), nb, cur_stream());
+  return slab;
 }
 
 torch::Tensor kmeans_phase_probe(torch::Tensor x, torch::Tensor centers,

@@ -173,8 +173,7 @@ template <int FAM, int PHASES = 7>
 __global__ __launch_bounds__(NT) void glm_value_grad_k(
     const float* __restrict__ x, const float* __restrict__ y,
     const float* __restrict__ w, float b, long m, int d, int ld, int nbuf,
-    double* __restrict__ loss_slab, double* __restrict__ gw_slab,
-    double* __restrict__ gb_slab) {
+    double* __restrict__ slab) {  // [nblocks, d+2]: gw | loss | gb
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* sw = (float*)smem_raw;                    // [d]
   float* xtbuf = sw + ((d + 3) & ~3);              // nbuf x [TR * ld]
@@ -318,21 +317,23 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
     lds_barrier();
     cur ^= dbuf ? 1 : 0;
   }
-  // block reduction of loss / gb
+  // block reduction of loss then gb through ONE red array (the second
+  // array cost 2 KB of LDS and the WG/CU occupancy step at d~100)
   red[tid] = lacc;
-  red[NT + tid] = gbacc;
   __syncthreads();
   for (int s = NT / 2; s > 0; s >>= 1) {
-    if (tid < s) {
-      red[tid] += red[tid + s];
-      red[NT + tid] += red[NT + tid + s];
-    }
+    if (tid < s) red[tid] += red[tid + s];
     __syncthreads();
   }
-  if (tid == 0) {
-    loss_slab[blockIdx.x] = red[0];
-    gb_slab[blockIdx.x] = red[NT];
+  if (tid == 0) slab[(long)blockIdx.x * (d + 2) + d] = red[0];
+  __syncthreads();
+  red[tid] = gbacc;
+  __syncthreads();
+  for (int s = NT / 2; s > 0; s >>= 1) {
+    if (tid < s) red[tid] += red[tid + s];
+    __syncthreads();
   }
+  if (tid == 0) slab[(long)blockIdx.x * (d + 2) + d + 1] = red[0];
   // combine the 4 waves' gradient partials through LDS (xtbuf reusable now)
   double* gred = (double*)xtbuf;  // [4 * d] doubles (d <= 512 -> 16 KB)
 #pragma unroll
@@ -342,7 +343,7 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
   }
   __syncthreads();
   for (int c = tid; c < d; c += NT)
-    gw_slab[(long)blockIdx.x * d + c] =
+    slab[(long)blockIdx.x * (d + 2) + c] =
         gred[c] + gred[d + c] + gred[2 * d + c] + gred[3 * d + c];
 }
 
@@ -748,32 +749,28 @@ void launch_col_minmax(const float* x, long m, int f, float* mins,
 
 void launch_glm_value_grad(const float* x, const float* y, const float* w,
                            float b, long m, int d, int family,
-                           double* loss_slab, double* gw_slab,
-                           double* gb_slab, int nblocks, hipStream_t stream) {
+                           double* slab, int nblocks, hipStream_t stream) {
   // d%4==0: contiguous LDS image (float4 staging, <=2-way bank conflicts);
   // otherwise odd stride (conflict-free, scalar staging).
   const int ld = ((d & 3) == 0) ? d : (d | 1);
   auto smem_for = [&](int nbuf) {
     return ((((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
-              sizeof(float) + 15) & ~15)) + 2 * NT * sizeof(double);
+              sizeof(float) + 15) & ~15)) + NT * sizeof(double);
   };
   const int nbuf = (smem_for(2) <= 158 * 1024) ? 2 : 1;
   const size_t smem = smem_for(nbuf);
   switch (family) {
     case 0:
       hipLaunchKernelGGL((glm_value_grad_k<0>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, nbuf, loss_slab, gw_slab,
-                         gb_slab);
+                         stream, x, y, w, b, m, d, ld, nbuf, slab);
       break;
     case 1:
       hipLaunchKernelGGL((glm_value_grad_k<1>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, nbuf, loss_slab, gw_slab,
-                         gb_slab);
+                         stream, x, y, w, b, m, d, ld, nbuf, slab);
       break;
     default:
       hipLaunchKernelGGL((glm_value_grad_k<2>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, nbuf, loss_slab, gw_slab,
-                         gb_slab);
+                         stream, x, y, w, b, m, d, ld, nbuf, slab);
   }
 }
 
@@ -790,18 +787,16 @@ void launch_kmeans_assign(const float* x, const float* cen, long m, int d,
 }
 
 void launch_glm_probe(const float* x, const float* y, const float* w,
-                      float b, long m, int d, int phases, double* loss_slab,
-                      double* gw_slab, double* gb_slab, int nblocks,
-                      hipStream_t stream) {
+                      float b, long m, int d, int phases, double* slab,
+                      int nblocks, hipStream_t stream) {
   const int ld = ((d & 3) == 0) ? d : (d | 1);
   const size_t smem =
       ((((((d + 3) & ~3) + 2 * GLM_TR * ld + 3 * GLM_TR) * sizeof(float) +
-         15) & ~15)) + 2 * NT * sizeof(double);
+         15) & ~15)) + NT * sizeof(double);
 #define GP(PH)                                                              \
   case PH:                                                                  \
     hipLaunchKernelGGL((glm_value_grad_k<0, PH>), dim3(nblocks), dim3(NT),  \
-                       smem, stream, x, y, w, b, m, d, ld, 2, loss_slab,    \
-                       gw_slab, gb_slab);                                   \
+                       smem, stream, x, y, w, b, m, d, ld, 2, slab);        \
     break
   switch (phases) { GP(0); GP(1); GP(3); GP(5); GP(7); }
 #undef GP

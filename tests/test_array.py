@@ -131,3 +131,40 @@ def test_hist_world2(mp_runner):
     rng = np.random.default_rng(12)
     a = rng.standard_normal(1001).astype(np.float32)
     mp_runner(_hist_world2, world=2, args=(a.tobytes(), len(a)))
+
+
+def _tiny_world2(rank, world):
+    import torch
+    from mi355ml.preprocessing import StandardScaler
+    from mi355ml.cluster import KMeans
+    from mi355ml.linear_model import LogisticRegression
+    # 3 rows over 2 ranks: rank 1 gets 1 row; 1-row arrays everywhere
+    a = np.array([[1.0, 2.0], [3.0, 4.0], [5.0, 6.0]], dtype=np.float32)
+    y = np.array([0.0, 1.0, 1.0], dtype=np.float32)
+    X = from_numpy(a)
+    ss = StandardScaler().fit(X)
+    np.testing.assert_allclose(ss.mean_.cpu().numpy(), a.mean(0), rtol=1e-6)
+    lr = LogisticRegression(C=1.0, max_iter=20).fit(X, from_numpy(y))
+    assert lr.coef_.shape == (2,)
+    km = KMeans(n_clusters=2, random_state=0, max_iter=5).fit(X)
+    assert km.cluster_centers_.shape == (2, 2)
+
+
+def test_tiny_data_world2(mp_runner):
+    mp_runner(_tiny_world2, world=2)
+
+
+def _empty_rank_world2(rank, world):
+    # fewer rows than ranks would give rank 1 zero rows at world 3; at
+    # world 2 use a 1-row array so rank 1 is empty
+    a = np.array([[7.0, 8.0, 9.0]], dtype=np.float32)
+    X = from_numpy(a)
+    if rank == 1:
+        assert X.local.shape[0] == 0
+    np.testing.assert_allclose(X.mean().cpu().numpy(), a.mean(0), rtol=1e-6)
+    np.testing.assert_allclose(X.min().cpu().numpy(), a.min(0))
+    np.testing.assert_array_equal(X.gather().cpu().numpy(), a)
+
+
+def test_empty_rank_world2(mp_runner):
+    mp_runner(_empty_rank_world2, world=2)
