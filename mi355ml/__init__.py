@@ -16,13 +16,13 @@ from .core.array import DistArray, from_numpy, shard_rows
 from .core.bag import DistList
 from .parallel import comm
 
-from . import (cluster, compose, datasets, decomposition, diagnostics, ensemble,
+from . import (cluster, compose, dataframe, datasets, decomposition, diagnostics, ensemble,
                feature_extraction, impute, linear_model, metrics,
                model_selection, naive_bayes, preprocessing, utils, wrappers)
 
 __all__ = [
     "config", "comm", "DistArray", "DistList", "from_numpy", "shard_rows",
-    "cluster", "compose", "datasets", "decomposition", "diagnostics", "ensemble",
+    "cluster", "compose", "dataframe", "datasets", "decomposition", "diagnostics", "ensemble",
     "feature_extraction", "impute", "linear_model", "metrics",
     "model_selection", "naive_bayes", "preprocessing", "utils", "wrappers",
     "__version__",

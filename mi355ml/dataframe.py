@@ -1,0 +1,98 @@
+"""Partitioned-dataframe helpers (the used subset of the reference's
+dask.dataframe layer — SURVEY §2.3 notes only df-preprocessing is needed;
+Categorizer/DummyEncoder live in preprocessing._dataframe).
+
+Frames are pandas DataFrames partitioned over ranks in a DistList. The
+groupby aggregation follows the reference's ACA (apply-concat-apply)
+groupby: per-partition partial aggregates, ONE ragged object all-gather,
+and a deterministic combine — mean carries (sum, count) partials so the
+result is exact, never an average of averages.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional, Union
+
+import numpy as np
+import pandas as pd
+
+from .core.bag import DistList
+
+
+def from_pandas(df: pd.DataFrame, npartitions: Optional[int] = None
+                ) -> DistList:
+    """Split a DataFrame into row partitions sharded over ranks."""
+    from .parallel import comm as comm_mod
+    comm = comm_mod.default()
+    nparts = npartitions or comm.world_size
+    bounds = np.linspace(0, len(df), nparts + 1).astype(int)
+    parts = [df.iloc[bounds[i]:bounds[i + 1]] for i in range(nparts)]
+    return DistList.from_global(parts)
+
+
+def map_partitions(fn: Callable, frames: DistList) -> DistList:
+    """Apply fn to every partition (dask map_partitions analog)."""
+    return frames.map(fn)
+
+
+def concat(frames: DistList) -> pd.DataFrame:
+    """Materialize the full frame on every rank."""
+    parts = frames.gather()
+    return pd.concat(parts) if parts else pd.DataFrame()
+
+
+def groupby_agg(frames: DistList, by: Union[str, List[str]],
+                agg: str = "sum",
+                columns: Optional[List[str]] = None) -> pd.DataFrame:
+    """Distributed groupby aggregate: agg in {sum, count, mean, min, max}.
+
+    Returns the aggregated frame, replicated on every rank.
+    """
+    if agg not in ("sum", "count", "mean", "min", "max"):
+        raise ValueError(f"unsupported agg {agg!r}")
+
+    def _partial(part: pd.DataFrame):
+        if part is None or not len(part):
+            return None
+        sub = part if columns is None else part[
+            ([by] if isinstance(by, str) else list(by)) + list(columns)]
+        g = sub.groupby(by)
+        if agg == "mean":  # carry (sum, count) — never average averages
+            return {"sum": g.sum(numeric_only=True),
+                    "count": g.count()}
+        if agg == "count":
+            return {"count": g.count()}
+        return {agg: getattr(g, agg)(numeric_only=(agg == "sum"))}
+
+    local = [_partial(p) for p in frames.local]
+    gathered = frames.comm.all_gather_object(local)
+    partials = [p for rank_parts in gathered for p in rank_parts
+                if p is not None]
+    if not partials:
+        return pd.DataFrame()
+
+    def _combine(key: str, how: str) -> pd.DataFrame:
+        cat = pd.concat([p[key] for p in partials])
+        g = cat.groupby(level=list(range(cat.index.nlevels)))
+        return getattr(g, how)()
+
+    if agg == "mean":
+        s = _combine("sum", "sum")
+        c = _combine("count", "sum")
+        return s / c
+    if agg in ("sum", "count"):
+        return _combine(agg if agg != "count" else "count", "sum")
+    return _combine(agg, agg)  # min of mins / max of maxes
+
+
+def value_counts(frames: DistList, column: str) -> pd.Series:
+    def _vc(part):
+        return part[column].value_counts() if len(part) else None
+
+    local = [_vc(p) for p in frames.local]
+    gathered = frames.comm.all_gather_object(local)
+    parts = [p for rp in gathered for p in rp if p is not None]
+    if not parts:
+        return pd.Series(dtype=np.int64)
+    return pd.concat(parts).groupby(level=0).sum().sort_values(
+        ascending=False)

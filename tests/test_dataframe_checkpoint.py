@@ -72,3 +72,46 @@ def test_native_glm_partial_fit_incremental():
     for _ in range(30):
         lr.partial_fit(from_numpy(X), from_numpy(y))
     assert lr.score(from_numpy(X), from_numpy(y)) > 0.85
+
+
+def test_dataframe_groupby_agg():
+    from mi355ml import dataframe as mdf
+    rng = np.random.default_rng(7)
+    df = pd.DataFrame({
+        "g": rng.choice(["a", "b", "c"], 1000),
+        "v": rng.standard_normal(1000),
+        "w": rng.integers(0, 10, 1000),
+    })
+    frames = mdf.from_pandas(df, npartitions=4)
+    for agg in ("sum", "count", "mean", "min", "max"):
+        got = mdf.groupby_agg(frames, "g", agg=agg, columns=["v", "w"])
+        if agg == "count":
+            ref = df.groupby("g")[["v", "w"]].count()
+        elif agg == "sum":
+            ref = df.groupby("g")[["v", "w"]].sum(numeric_only=True)
+        else:
+            ref = getattr(df.groupby("g")[["v", "w"]], agg)(
+                numeric_only=(agg == "sum"))
+        pd.testing.assert_frame_equal(got.sort_index(), ref.sort_index(),
+                                      check_dtype=False, check_like=True)
+    vc = mdf.value_counts(frames, "g")
+    pd.testing.assert_series_equal(
+        vc.sort_index(), df["g"].value_counts().sort_index(),
+        check_names=False, check_dtype=False)
+    full = mdf.concat(frames)
+    assert len(full) == 1000
+
+
+def _df_world2(rank, world):
+    from mi355ml import dataframe as mdf
+    df = pd.DataFrame({"g": ["x", "y"] * 50, "v": np.arange(100.0)})
+    frames = mdf.from_pandas(df, npartitions=2)
+    assert len(frames.local) == 1
+    got = mdf.groupby_agg(frames, "g", agg="mean", columns=["v"])
+    ref = df.groupby("g")[["v"]].mean()
+    pd.testing.assert_frame_equal(got.sort_index(), ref.sort_index(),
+                                  check_dtype=False)
+
+
+def test_dataframe_world2(mp_runner):
+    mp_runner(_df_world2, world=2)
