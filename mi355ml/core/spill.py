@@ -117,6 +117,23 @@ class SpilledDistArray(DistArray):
         return self._vec(out)
 
 
+    # -- transforms stream through the device ----------------------------
+    def map_local(self, fn):
+        """Apply fn block-by-block THROUGH the device (H2D -> fn -> D2H),
+        result pinned host again — scalers/transforms on out-of-HBM data
+        run on the GPU instead of host torch."""
+        outs = []
+        for _, blk in self.device_blocks():
+            res = fn(blk)
+            outs.append(res.cpu())
+        local = (torch.cat(outs, dim=0) if outs else
+                 fn(self.local[:0]).cpu())
+        if self.comm.device.type == "cuda":
+            local = local.pin_memory()
+        return SpilledDistArray(local, self.global_rows, self.row_offset,
+                                self.comm, self.block_rows)
+
+
 def spill_rows(full: torch.Tensor, comm=None,
                block_rows: Optional[int] = None,
                pin: bool = True) -> SpilledDistArray:

@@ -57,3 +57,16 @@ def test_spilled_kmeans():
     d = np.linalg.norm(c1[:, None, :] - c2[None, :, :], axis=2)
     assert d.min(axis=1).max() < 0.3, d.min(axis=1)
     assert abs(km1.inertia_ - km2.inertia_) / km1.inertia_ < 5e-3
+
+
+def test_spilled_transform_streams():
+    from mi355ml.preprocessing import StandardScaler
+    a = _data(3000, 4, seed=3)
+    S = spill_rows(torch.from_numpy(a), block_rows=700)
+    sc = StandardScaler().fit(S)
+    out = sc.transform(S)
+    from mi355ml.core.spill import SpilledDistArray
+    assert isinstance(out, SpilledDistArray)
+    got = out.gather().cpu().numpy()
+    ref = (a - a.mean(0)) / a.std(0)
+    np.testing.assert_allclose(got, ref, rtol=1e-3, atol=1e-4)
