@@ -78,7 +78,6 @@ class _SearchBase(BaseEstimator):
             Xtr_c, Xte_c = Xtr, Xte
             prefix_token = str(fold)
             steps = est.steps
-            fitted_steps = []
             for name, step in steps[:-1]:
                 prefix_token = _tokenize([prefix_token, name, step])
                 hit = cache.get(prefix_token)
@@ -88,7 +87,6 @@ class _SearchBase(BaseEstimator):
                     Xte_n = stf.transform(Xte_c)
                     cache[prefix_token] = (stf, Xtr_n, Xte_n)
                 stf, Xtr_c, Xte_c = cache[prefix_token]
-                fitted_steps.append((name, stf))
             final_name, final = steps[-1]
             finalf = clone(final)
             if ytr is not None:

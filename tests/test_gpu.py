@@ -217,3 +217,18 @@ def test_glm_wide_features_gpu():
     l_, gw, gb = glm_value_grad(x.cuda(), y.cuda(), w.cuda(), b, "logistic")
     torch.testing.assert_close(l_.cpu(), l_ref, rtol=1e-4, atol=1e-2)
     torch.testing.assert_close(gw.cpu(), gw_ref, rtol=1e-3, atol=1e-2)
+
+
+def test_glm_gpu_bitwise_deterministic():
+    """Partial-slab design (no atomics): bitwise run-to-run."""
+    from mi355ml.ops import glm_value_grad
+    g = torch.Generator().manual_seed(31)
+    x = torch.randn(500_000, 100, generator=g).cuda()
+    y = (torch.rand(500_000, generator=g) < 0.5).float().cuda()
+    w = (torch.randn(100, generator=g) * 0.1).cuda()
+    b = torch.tensor(0.2)
+    r1 = glm_value_grad(x, y, w, b, "logistic")
+    r2 = glm_value_grad(x, y, w, b, "logistic")
+    assert torch.equal(r1[0], r2[0])
+    assert torch.equal(r1[1], r2[1])
+    assert torch.equal(r1[2], r2[2])
