@@ -48,10 +48,10 @@ def merge_percentiles(qs: Sequence[float], curves: torch.Tensor,
     cum = counts_s.cumsum(dim=0)
     total = ns.sum()
     targets = torch.tensor(qs, dtype=torch.float64, device=dev) / 100.0 * total
-    out = torch.empty(len(qs), F, dtype=torch.float64, device=dev)
-    for f in range(F):
-        out[:, f] = _interp(targets, cum[:, f], vals_s[:, f])
-    return out
+    # batched per-column interpolation (one searchsorted over [F, G*nq])
+    out = _interp_batched(targets.expand(F, -1), cum.t().contiguous(),
+                          vals_s.t().contiguous())
+    return out.t().contiguous()
 
 
 def _interp(xq: torch.Tensor, xp: torch.Tensor, fp: torch.Tensor
@@ -68,6 +68,23 @@ def _interp(xq: torch.Tensor, xp: torch.Tensor, fp: torch.Tensor
     res = y0 + t.clamp(0, 1) * (y1 - y0)
     res = torch.where(xq <= xp[0], fp[0], res)
     res = torch.where(xq >= xp[-1], fp[-1], res)
+    return res
+
+
+def _interp_batched(xq: torch.Tensor, xp: torch.Tensor, fp: torch.Tensor
+                    ) -> torch.Tensor:
+    """Row-batched np.interp: xq [F, M], xp/fp [F, N] (each row monotone).
+    One batched searchsorted instead of a per-column Python loop."""
+    idx = torch.searchsorted(xp, xq.contiguous(), right=False)
+    idx = idx.clamp(1, xp.shape[1] - 1)
+    x0 = torch.gather(xp, 1, idx - 1)
+    x1 = torch.gather(xp, 1, idx)
+    y0 = torch.gather(fp, 1, idx - 1)
+    y1 = torch.gather(fp, 1, idx)
+    t = torch.where(x1 > x0, (xq - x0) / (x1 - x0), torch.zeros_like(xq))
+    res = y0 + t.clamp(0, 1) * (y1 - y0)
+    res = torch.where(xq <= xp[:, :1], fp[:, :1].expand_as(res), res)
+    res = torch.where(xq >= xp[:, -1:], fp[:, -1:].expand_as(res), res)
     return res
 
 

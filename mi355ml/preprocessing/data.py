@@ -189,19 +189,19 @@ class QuantileTransformer(BaseEstimator, TransformerMixin):
         return self
 
     def _map(self, t: torch.Tensor, forward: bool) -> torch.Tensor:
-        from ..ops.percentile import _interp
+        from ..ops.percentile import _interp_batched
         dev = t.device
-        q = self.quantiles_.to(dev)
-        refs = self.references_.to(dev)
-        out = torch.empty_like(t)
-        for f in range(t.shape[1]):
-            col = t[:, f].double()
-            if forward:
-                r = _interp(col, q[:, f].contiguous(), refs)
-            else:
-                r = _interp(col, refs.contiguous(), q[:, f])
-            out[:, f] = r.to(t.dtype)
-        return out
+        q = self.quantiles_.to(dev)                       # [nq, F]
+        refs = self.references_.to(dev)                   # [nq]
+        F = t.shape[1]
+        xq = t.t().double().contiguous()                  # [F, m]
+        if forward:
+            r = _interp_batched(xq, q.t().contiguous(),
+                                refs.expand(F, -1).contiguous())
+        else:
+            r = _interp_batched(xq, refs.expand(F, -1).contiguous(),
+                                q.t().contiguous())
+        return r.t().to(t.dtype).contiguous()
 
     def transform(self, X):
         check_is_fitted(self, "quantiles_")
