@@ -91,22 +91,45 @@ def _interp_batched(xq: torch.Tensor, xp: torch.Tensor, fp: torch.Tensor
 def col_percentiles(X: DistArray, qs: Sequence[float]) -> torch.Tensor:
     """Global per-column percentiles of a row-sharded array.
 
-    Returns [len(qs), F] float64, replicated.
+    Returns [len(qs), F] float64, replicated. Spilled shards contribute one
+    per-BLOCK curve each (the merge handles any number of curves — the same
+    approximation class as the reference's per-chunk summaries).
     """
-    x = X._as2d()
-    dev = x.device
+    from ..core.spill import SpilledDistArray
+    dev = X.device
     q_ext = sorted(set([0.0, 100.0] + [float(q) for q in qs]))
     q_ext_t = torch.tensor(q_ext, dtype=torch.float64)
-    curve = _shard_curve(x, q_ext_t)
-    n = torch.tensor([float(x.shape[0])], dtype=torch.float64, device=dev)
-    if X.comm.world_size > 1:
-        curves = torch.stack(X.comm.all_gather(curve.contiguous()))
-        ns = torch.cat(X.comm.all_gather(n))
-        # drop empty shards (NaN curves)
-        keep = ns > 0
-        curves, ns = curves[keep], ns[keep]
+    local_curves = []
+    local_ns = []
+    if isinstance(X, SpilledDistArray):
+        for _, blk in X.device_blocks():
+            b2 = blk if blk.dim() == 2 else blk[:, None]
+            local_curves.append(_shard_curve(b2, q_ext_t))
+            local_ns.append(float(b2.shape[0]))
     else:
-        curves, ns = curve[None], n
+        x = X._as2d()
+        local_curves.append(_shard_curve(x, q_ext_t))
+        local_ns.append(float(x.shape[0]))
+    # pad to a fixed per-rank curve count for the dense all-gather
+    nloc = torch.tensor([len(local_curves)], dtype=torch.int64, device=dev)
+    if X.comm.world_size > 1:
+        counts = [int(c.item()) for c in X.comm.all_gather(nloc)]
+        gmax = max(counts)
+        F = local_curves[0].shape[1]
+        nq = len(q_ext)
+        while len(local_curves) < gmax:
+            local_curves.append(torch.full((nq, F), float("nan"),
+                                           dtype=torch.float64, device=dev))
+            local_ns.append(0.0)
+        curves = torch.cat(X.comm.all_gather(
+            torch.stack(local_curves).contiguous()))
+        ns = torch.cat(X.comm.all_gather(
+            torch.tensor(local_ns, dtype=torch.float64, device=dev)))
+    else:
+        curves = torch.stack(local_curves)
+        ns = torch.tensor(local_ns, dtype=torch.float64, device=dev)
+    keep = ns > 0
+    curves, ns = curves[keep], ns[keep]
     merged = merge_percentiles([float(q) for q in qs], curves, ns,
-                               q_ext_t.to(dev))
+                               q_ext_t.to(curves.device))
     return merged

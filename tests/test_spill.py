@@ -70,3 +70,19 @@ def test_spilled_transform_streams():
     got = out.gather().cpu().numpy()
     ref = (a - a.mean(0)) / a.std(0)
     np.testing.assert_allclose(got, ref, rtol=1e-3, atol=1e-4)
+
+
+def test_spilled_percentiles_and_robust_scaler():
+    from mi355ml.ops.percentile import col_percentiles
+    from mi355ml.preprocessing import RobustScaler
+    a = _data(5000, 3, seed=5)
+    X = from_numpy(a)
+    S = spill_rows(torch.from_numpy(a), block_rows=800)
+    qs = col_percentiles(S, [25.0, 50.0, 75.0]).cpu().numpy()
+    ref = np.percentile(a, [25, 50, 75], axis=0)
+    np.testing.assert_allclose(qs, ref, rtol=0.05, atol=0.05)
+    rs = RobustScaler().fit(S)
+    rs2 = RobustScaler().fit(X)
+    np.testing.assert_allclose(rs.center_.cpu().numpy(),
+                               rs2.center_.cpu().numpy(), rtol=0.05,
+                               atol=0.05)
