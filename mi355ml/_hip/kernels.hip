@@ -387,13 +387,13 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
   float* xtbuf = nhc + k16;                      // 2 x-tile buffers
   float* xn = xtbuf + 2 * KM_TR * d4;            // row norms [KM_TR]
   int* rlab = (int*)(xn + KM_TR);                // row labels [KM_TR]
-  float* cbest = (float*)(rlab + KM_TR);         // per-half candidates
-  int* cbj = (int*)(cbest + 2 * KM_TR);          // [2][KM_TR]
-  float* bins = (float*)(cbj + 2 * KM_TR);       // UPDATE: [k*d]
+  float* cbest = (float*)(rlab + KM_TR);         // [16][KM_TR+1] candidates
+  int* cbj = (int*)(cbest + 16 * (KM_TR + 1));   // [16][KM_TR+1]
+  float* bins = (float*)(cbj + 16 * (KM_TR + 1));  // UPDATE: [k*d]
   float* bcnt = bins + (UPDATE ? k * d : 0);     // UPDATE: [k]
   double* red = (double*)(smem_raw + ((((size_t)(k16 * d4 + k16 +
-      2 * KM_TR * d4 + 6 * KM_TR + (UPDATE ? (k * d + k) : 0)) * 4 + 15) &
-      ~15)));
+      2 * KM_TR * d4 + 2 * KM_TR + 32 * (KM_TR + 1) +
+      (UPDATE ? (k * d + k) : 0)) * 4 + 15) & ~15)));
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;   // 0..7
@@ -555,52 +555,53 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
         asm volatile("" :: "v"(best0), "v"(best1), "v"(best2), "v"(best3),
                      "v"(bj0), "v"(bj1), "v"(bj2), "v"(bj3));
       } else {
-#define KM_RED(BEST, BJ)                                                   \
-  for (int off = 8; off > 0; off >>= 1) {                                  \
-    const float ov = __shfl_xor(BEST, off, 16);                            \
-    const int oj = __shfl_xor(BJ, off, 16);                                \
+      // ONE shfl halving (4 chained steps cost ~4k cycles/tile — 30% of
+      // the compute phase); the remaining 16 candidates per row go through
+      // a padded LDS publish and a batched row-owner finalize instead.
+#define KM_RED1(BEST, BJ)                                                  \
+  {                                                                        \
+    const float ov = __shfl_xor(BEST, 8, 16);                              \
+    const int oj = __shfl_xor(BJ, 8, 16);                                  \
     if (ov > BEST || (ov == BEST && oj < BJ)) { BEST = ov; BJ = oj; }      \
   }
-      KM_RED(best0, bj0)
-      KM_RED(best1, bj1)
-      KM_RED(best2, bj2)
-      KM_RED(best3, bj3)
-#undef KM_RED
-      // publish this half's per-row candidates, then half 0 finalizes
-      if (lj == 0) {
+      KM_RED1(best0, bj0)
+      KM_RED1(best1, bj1)
+      KM_RED1(best2, bj2)
+      KM_RED1(best3, bj3)
+#undef KM_RED1
+      if (lj < 8) {  // 8 surviving lanes per half publish their rows
+        const int slot = half * 8 + lj;
         const int rbase = qw * 16 + lk * 4;
         const float bests[4] = {best0, best1, best2, best3};
         const int bjs[4] = {bj0, bj1, bj2, bj3};
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
-          const int r = rbase + g;
-          cbest[half * KM_TR + r] = bests[g];
-          cbj[half * KM_TR + r] = bjs[g];
+          cbest[slot * (KM_TR + 1) + rbase + g] = bests[g];
+          cbj[slot * (KM_TR + 1) + rbase + g] = bjs[g];
         }
       }
       lds_barrier();
-      if (half == 0 && lj == 0) {
-        const int rbase = qw * 16 + lk * 4;
+      if (tid < KM_TR && tid < rows) {  // wave 0: one lane per row
+        const int r = tid;
+        float bv = cbest[r];
+        int bj = cbj[r];
+        const int nslots = T1 > 0 ? 16 : 8;
 #pragma unroll
-        for (int g = 0; g < 4; ++g) {
-          const int r = rbase + g;
-          if (r < rows) {
-            float bv = cbest[r];
-            int bj = cbj[r];
-            const float ov = cbest[KM_TR + r];
-            const int oj = cbj[KM_TR + r];
-            if (T1 > 0 && (ov > bv || (ov == bv && oj < bj))) {
-              bv = ov;
-              bj = oj;
-            }
-            labels[r0 + r] = bj;
-            if (UPDATE) {
-              rlab[r] = bj;
-              iacc += -2.0 * (double)bv;  // + Σ‖x‖² from the update pass
-            } else {
-              dists[r0 + r] = fmaxf(xn[r] - 2.0f * bv, 0.0f);
-            }
+        for (int sl = 1; sl < 16; ++sl) {
+          if (sl >= nslots) break;
+          const float ov = cbest[sl * (KM_TR + 1) + r];
+          const int oj = cbj[sl * (KM_TR + 1) + r];
+          if (ov > bv || (ov == bv && oj < bj)) {
+            bv = ov;
+            bj = oj;
           }
+        }
+        labels[r0 + r] = bj;  // 64-lane coalesced store
+        if (UPDATE) {
+          rlab[r] = bj;
+          iacc += -2.0 * (double)bv;  // + Σ‖x‖² from the update pass
+        } else {
+          dists[r0 + r] = fmaxf(xn[r] - 2.0f * bv, 0.0f);
         }
       }
       }  // PHASES & 8 (epilogue)
@@ -704,7 +705,8 @@ extern "C" __global__ __launch_bounds__(NT) void affine_transform_k(
 static size_t kmeans_smem_bytes_impl(int k, int d, int update) {
   const int d4 = (d + 3) & ~3;
   const int k16 = (k + 15) & ~15;
-  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)2 * 64 * d4 + 6 * 64;
+  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)2 * 64 * d4 + 2 * 64 +
+                32 * 65;
   if (update) f32s += (size_t)k * d + k;
   return (((f32s * 4 + 15) & ~15)) + 512 * sizeof(double);
 }

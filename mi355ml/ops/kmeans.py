@@ -30,7 +30,7 @@ def _fits_fused(k: int, d: int, update: bool) -> bool:
         return False
     d4 = (d + 3) & ~3
     k16 = (k + 15) & ~15
-    f32s = k16 * d4 + k16 + 2 * 64 * d4 + 6 * 64
+    f32s = k16 * d4 + k16 + 2 * 64 * d4 + 2 * 64 + 32 * 65
     if update:
         f32s += k * d + k
     return f32s * 4 + 16 + 512 * 8 <= 160 * 1024 - 1024
