@@ -51,7 +51,13 @@ def _local_qr(a: torch.Tensor, d: int, keep_q: bool
         q, r = torch.linalg.qr(torch.cat([a, pad], dim=0), mode="reduced")
         return (q[:m] if keep_q else None), r
     try:
-        g = a.t().double() @ a.double()
+        # Gram in the INPUT dtype (rocBLAS f64 GEMM runs at ~20 GFLOPS on
+        # these thin shapes — 70x slower than sgemm); only the tiny [d, d]
+        # result is upcast for the Cholesky. CholeskyQR2's second pass
+        # restores Q orthogonality; a ~sqrt(n)*eps_f32 Gram perturbation
+        # moves singular values ~1e-4 relative, inside this framework's
+        # fp32 tolerance class.
+        g = (a.t() @ a).double()
         r1 = torch.linalg.cholesky(g, upper=True)  # XᵀX = R1ᵀ R1
         if not keep_q:
             return None, r1
@@ -60,7 +66,7 @@ def _local_qr(a: torch.Tensor, d: int, keep_q: bool
             upper=True)
         q1 = a @ rinv1.to(a.dtype)
         # second pass: re-orthogonalize (CholeskyQR2)
-        g2 = q1.t().double() @ q1.double()
+        g2 = (q1.t() @ q1).double()
         r2 = torch.linalg.cholesky(g2, upper=True)
         rinv2 = torch.linalg.solve_triangular(
             r2, torch.eye(d, dtype=torch.float64, device=a.device),
