@@ -166,6 +166,13 @@ def init(backend: Optional[str] = None) -> Communicator:
             backend = config.get("comm.backend")
         if backend == "auto":
             backend = "nccl" if device.type == "cuda" else "gloo"
+        if backend == "gloo" and device.type == "cuda":
+            import warnings
+            warnings.warn(
+                "comm backend 'gloo' with a CUDA device: gloo collectives "
+                "do not support device tensors — use backend 'nccl' (RCCL) "
+                "on GPUs, or hide the GPUs for CPU-only tests.",
+                RuntimeWarning, stacklevel=2)
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29531")
         dist.init_process_group(

@@ -33,13 +33,25 @@ def _to_plain(res) -> List:
     return np.asarray(res).tolist()
 
 
-def create_app(model: Any):
+def create_app(model: Any, max_batch: int = 65536):
     from starlette.applications import Starlette
-    from starlette.responses import JSONResponse
+    from starlette.responses import JSONResponse, PlainTextResponse
     from starlette.routing import Route
+
+    from .diagnostics import COUNTERS, device_memory
 
     async def health(request):
         return JSONResponse({"status": "ok"})
+
+    async def metrics(request):
+        """Prometheus text exposition of the process counters + HBM usage."""
+        lines = []
+        for k, v in sorted(COUNTERS.snapshot().items()):
+            name = "mi355ml_" + k.replace(".", "_").replace("-", "_")
+            lines.append(f"{name} {v}")
+        for k, v in device_memory().items():
+            lines.append(f"mi355ml_device_{k} {v}")
+        return PlainTextResponse("\n".join(lines) + "\n")
 
     async def info(request):
         return JSONResponse({
@@ -67,16 +79,25 @@ def create_app(model: Any):
                 return JSONResponse({"detail": "data must be a 2-D float "
                                                "array under key 'data'"},
                                     status_code=422)
+            if x.shape[0] > max_batch:
+                return JSONResponse(
+                    {"detail": f"batch of {x.shape[0]} rows exceeds the "
+                               f"server limit of {max_batch}"},
+                    status_code=413)
             try:
                 res = getattr(model, name)(x)
             except Exception as e:  # surface model errors as 400s
+                COUNTERS.inc("serving.errors")
                 return JSONResponse({"detail": str(e)}, status_code=400)
+            COUNTERS.inc("serving.requests")
+            COUNTERS.inc("serving.rows", float(x.shape[0]))
             return JSONResponse({"result": _to_plain(res)})
 
         return endpoint
 
     return Starlette(routes=[
         Route("/health", health, methods=["GET"]),
+        Route("/metrics", metrics, methods=["GET"]),
         Route("/info", info, methods=["GET"]),
         Route("/predict", _method("predict"), methods=["POST"]),
         Route("/predict_proba", _method("predict_proba"), methods=["POST"]),
@@ -94,9 +115,12 @@ def main(argv: Optional[list] = None):
     ap.add_argument("model", help="joblib file with a fitted estimator")
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--max-batch", type=int, default=65536,
+                    help="largest accepted request batch (rows)")
     args = ap.parse_args(argv)
     model = joblib.load(args.model)
-    uvicorn.run(create_app(model), host=args.host, port=args.port)
+    uvicorn.run(create_app(model, max_batch=args.max_batch),
+                host=args.host, port=args.port)
 
 
 if __name__ == "__main__":

@@ -30,3 +30,20 @@ def test_serving_roundtrip():
     # malformed input -> 4xx, not a crash
     r = client.post("/predict", json={"data": "nope"})
     assert r.status_code == 422
+    # metrics: request/row counters visible in prometheus text format
+    m = client.get("/metrics")
+    assert m.status_code == 200
+    assert "mi355ml_serving_requests" in m.text
+    assert "mi355ml_serving_rows" in m.text
+
+
+def test_serving_batch_limit():
+    rng = np.random.default_rng(1)
+    X = rng.standard_normal((200, 3)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    lr = LogisticRegression(max_iter=20).fit(from_numpy(X), from_numpy(y))
+    client = TestClient(create_app(lr, max_batch=50))
+    r = client.post("/predict", json={"data": X[:51].tolist()})
+    assert r.status_code == 413
+    r = client.post("/predict", json={"data": X[:50].tolist()})
+    assert r.status_code == 200
