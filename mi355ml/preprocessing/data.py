@@ -118,18 +118,20 @@ class RobustScaler(BaseEstimator, TransformerMixin):
     percentiles (ops/percentile.py, SURVEY A.5 merge math)."""
 
     def __init__(self, with_centering: bool = True, with_scaling: bool = True,
-                 quantile_range=(25.0, 75.0), copy: bool = True):
+                 quantile_range=(25.0, 75.0), copy: bool = True,
+                 method: str = "default"):
         self.with_centering = with_centering
         self.with_scaling = with_scaling
         self.quantile_range = quantile_range
         self.copy = copy
+        self.method = method  # "default" (curve merge) or "tdigest"
 
     def fit(self, X, y=None):
         from ..ops.percentile import col_percentiles
         X = check_array(X)
         self.n_features_in_ = X.ncols
         lo, hi = self.quantile_range
-        qs = col_percentiles(X, [lo, 50.0, hi])
+        qs = col_percentiles(X, [lo, 50.0, hi], method=self.method)
         dt = X.dtype
         self.center_ = qs[1].to(dt) if self.with_centering else None
         if self.with_scaling:
@@ -170,12 +172,13 @@ class QuantileTransformer(BaseEstimator, TransformerMixin):
     def __init__(self, n_quantiles: int = 1000,
                  output_distribution: str = "uniform",
                  subsample: int = 100_000, random_state=None,
-                 copy: bool = True):
+                 copy: bool = True, method: str = "default"):
         self.n_quantiles = n_quantiles
         self.output_distribution = output_distribution
         self.subsample = subsample
         self.random_state = random_state
         self.copy = copy
+        self.method = method  # "default" (curve merge) or "tdigest"
 
     def fit(self, X, y=None):
         from ..ops.percentile import col_percentiles
@@ -184,7 +187,8 @@ class QuantileTransformer(BaseEstimator, TransformerMixin):
         refs = torch.linspace(0, 100, nq, dtype=torch.float64)
         self.n_quantiles_ = nq
         self.references_ = (refs / 100.0)
-        self.quantiles_ = col_percentiles(X, refs.tolist())  # [nq, F] f64
+        self.quantiles_ = col_percentiles(X, refs.tolist(),
+                                          method=self.method)  # [nq, F] f64
         self.n_features_in_ = X.ncols
         return self
 

@@ -137,3 +137,63 @@ def test_gaussian_nb():
     assert abs(acc - ref.score(X, y)) < 0.01
     proba = ours.predict_proba(from_numpy(X)).gather().cpu().numpy()
     np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-5)
+
+
+# -- t-digest percentile sketch (method='tdigest' parity) -------------------
+
+def test_tdigest_percentiles_accuracy():
+    rng = np.random.default_rng(3)
+    a = np.stack([rng.standard_normal(100_000),
+                  rng.exponential(3.0, 100_000),
+                  rng.uniform(-5, 5, 100_000)], axis=1).astype(np.float32)
+    from mi355ml.ops.percentile import col_percentiles
+    qs = [0.1, 1, 25, 50, 75, 99, 99.9]
+    td = col_percentiles(from_numpy(a), qs, method="tdigest").cpu().numpy()
+    # rank accuracy: each estimate must land within 0.1 percentile points
+    for c in range(a.shape[1]):
+        col = np.sort(a[:, c])
+        ranks = np.searchsorted(col, td[:, c]) / len(col) * 100
+        assert np.abs(ranks - np.asarray(qs)).max() < 0.1
+    # monotone in q, exact at the endpoints
+    assert (np.diff(td, axis=0) >= 0).all()
+    ends = col_percentiles(from_numpy(a), [0.0, 100.0],
+                           method="tdigest").cpu().numpy()
+    np.testing.assert_allclose(ends[0], a.min(axis=0), rtol=1e-6)
+    np.testing.assert_allclose(ends[1], a.max(axis=0), rtol=1e-6)
+
+
+def test_tdigest_nan_and_scalers():
+    rng = np.random.default_rng(4)
+    a = _data(20_000, 3, seed=4)
+    a[rng.integers(0, len(a), 300), 1] = np.nan
+    from mi355ml.ops.percentile import col_percentiles
+    td = col_percentiles(from_numpy(a), [50.0], method="tdigest")
+    med = np.nanmedian(a, axis=0)
+    np.testing.assert_allclose(td[0].cpu().numpy(), med, atol=0.05)
+    # RobustScaler(method='tdigest') agrees with sklearn
+    b = _data(seed=6)
+    ours = RobustScaler(method="tdigest").fit(from_numpy(b))
+    ref = SkRobust().fit(b)
+    np.testing.assert_allclose(ours.center_.cpu().numpy(), ref.center_,
+                               rtol=0.05, atol=0.05)
+    np.testing.assert_allclose(ours.scale_.cpu().numpy(), ref.scale_,
+                               rtol=0.05)
+    qt = QuantileTransformer(n_quantiles=200,
+                             method="tdigest").fit(from_numpy(b))
+    got = qt.transform(from_numpy(b)).gather().cpu().numpy()
+    assert np.allclose(got.mean(axis=0), 0.5, atol=0.02)
+
+
+def _tdigest_world2(rank, world, a_bytes, shape):
+    from mi355ml.ops.percentile import col_percentiles
+    a = np.frombuffer(a_bytes, dtype=np.float32).reshape(shape).copy()
+    qs = [5.0, 50.0, 95.0]
+    td = col_percentiles(from_numpy(a), qs, method="tdigest").cpu().numpy()
+    ex = np.percentile(a, qs, axis=0)
+    spread = a.max(axis=0) - a.min(axis=0)
+    assert (np.abs(td - ex) / spread).max() < 0.01
+
+
+def test_tdigest_world2(mp_runner):
+    a = _data(30_000, 4, seed=7)
+    mp_runner(_tdigest_world2, world=2, args=(a.tobytes(), a.shape))
