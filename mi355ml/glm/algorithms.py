@@ -30,7 +30,11 @@ from .regularizers import get_regularizer
 def _shard_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
                       family: str):
     """Per-shard fused (loss, gw, gb); streams blocks for spilled shards."""
+    from ..core.sparse import SparseDistArray
     from ..core.spill import SpilledDistArray
+    if isinstance(X, SparseDistArray):
+        from ..ops.glm import sparse_value_grad
+        return sparse_value_grad(X, y.local, w, b, family)
     if isinstance(X, SpilledDistArray):
         loss = gw = gb = None
         for lo, blk in X.device_blocks():

@@ -99,6 +99,9 @@ class _GLMBase(BaseEstimator):
         X = check_array(X)
         w = self.coef_.to(X.device)
         b = self.intercept_
+        from ..core.sparse import SparseDistArray
+        if isinstance(X, SparseDistArray):
+            return X.dot(w).map_local(lambda z: z + b)
         return X.map_local(lambda t: t @ w.to(t.dtype) + b)
 
     def predict(self, X):

@@ -66,6 +66,29 @@ def glm_value_grad(x: torch.Tensor, y: torch.Tensor, w: torch.Tensor,
     return loss, gw, gb
 
 
+def sparse_value_grad(X, y_local: torch.Tensor, w: torch.Tensor, b: float,
+                      family: str):
+    """Per-shard (loss, gw, gb) for a SparseDistArray shard (CSR SpMV pair,
+    f64 link math — the encoders' sparse one-hot output feeds the GLM
+    solvers without densifying)."""
+    z = X.dot(w.to(torch.float64)).local.double() + float(b)
+    yd = y_local.to(device=z.device, dtype=torch.float64)
+    if family == "logistic":
+        loss = (torch.nn.functional.softplus(z) - yd * z).sum()
+        r = torch.sigmoid(z) - yd
+    elif family == "normal":
+        diff = z - yd
+        loss = 0.5 * (diff * diff).sum()
+        r = diff
+    else:  # poisson
+        ez = torch.exp(z)
+        loss = (ez - yd * z).sum()
+        r = ez - yd
+    gw = X._xt_dot_local(r, torch.float64)
+    gb = r.sum()
+    return loss, gw, gb
+
+
 def _composed_value_grad(x, y, w, b, family):
     z = x @ w + b
     if family == "logistic":

@@ -85,6 +85,8 @@ class OneHotEncoder(_BaseEncoder):
         offsets = np.concatenate([[0], np.cumsum(sizes)])
         total = int(offsets[-1])
         dt = self.dtype
+        if self.sparse_output:
+            return self._transform_sparse(X, offsets, total)
 
         def _oh(t):
             codes = self._codes(t)
@@ -97,6 +99,25 @@ class OneHotEncoder(_BaseEncoder):
             return out
 
         return X.map_local(_oh)
+
+    def _transform_sparse(self, X, offsets, total):
+        """CSR shards straight from the codes — the dense one-hot matrix is
+        never materialized (dask-ml sparse_output=True parity)."""
+        from ..core.sparse import SparseDistArray
+        t = X._as2d()
+        m, F = t.shape
+        dev = t.device
+        codes = self._codes(t)                                   # [m, F]
+        off = torch.as_tensor(offsets[:-1], device=dev)
+        valid = codes >= 0
+        nnz_per_row = valid.sum(dim=1)
+        crow = torch.zeros(m + 1, dtype=torch.int64, device=dev)
+        crow[1:] = nnz_per_row.cumsum(0)
+        cols_full = codes + off[None, :]
+        col = cols_full[valid]                # row-major order = CSR order
+        vals = torch.ones(col.shape[0], dtype=self.dtype, device=dev)
+        return SparseDistArray(crow, col, vals, total, X.global_rows,
+                               X.row_offset, X.comm)
 
     def get_feature_names_out(self, input_features=None):
         names = (list(input_features) if input_features is not None

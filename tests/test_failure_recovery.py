@@ -62,6 +62,7 @@ def test_elastic_restart_resumes_from_checkpoint(tmp_path):
                         f"n_iter={{km.n_iter_}} inertia={{km.inertia_}}")
         comm_mod.shutdown()
     """))
+    from conftest import free_port
     env = dict(os.environ)
     env["CUDA_VISIBLE_DEVICES"] = ""
     env["HIP_VISIBLE_DEVICES"] = ""
@@ -70,7 +71,7 @@ def test_elastic_restart_resumes_from_checkpoint(tmp_path):
     res = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr=127.0.0.1",
-         "--master-port=29641", "--max-restarts=2",
+         f"--master-port={free_port()}", "--max-restarts=2",
          "--monitor-interval=1", str(script)],
         env=env, capture_output=True, text=True, timeout=420)
     assert done.exists(), (res.stdout[-2000:], res.stderr[-2000:])
