@@ -115,3 +115,38 @@ def test_kmeans_gpu_bitwise_deterministic():
     assert torch.equal(s1, s2)
     assert torch.equal(c1, c2)
     assert torch.equal(i1, i2)
+
+
+def test_tdigest_gpu():
+    from mi355ml.core.array import from_numpy
+    from mi355ml.ops.percentile import col_percentiles
+    rng = np.random.default_rng(11)
+    a = rng.standard_normal((500_000, 8)).astype(np.float32)
+    X = from_numpy(a)
+    assert X.device.type == "cuda"
+    qs = [1.0, 25.0, 50.0, 75.0, 99.0]
+    td = col_percentiles(X, qs, method="tdigest").cpu().numpy()
+    ex = np.percentile(a, qs, axis=0)
+    assert np.abs(td - ex).max() < 0.01
+
+
+def test_sparse_onehot_logreg_gpu():
+    import scipy.sparse as sp
+    from mi355ml.core.array import from_numpy
+    from mi355ml.core.sparse import SparseDistArray
+    from mi355ml.linear_model import LogisticRegression
+    from mi355ml.preprocessing import OneHotEncoder
+    rng = np.random.default_rng(12)
+    a = rng.integers(0, 8, (200_000, 5)).astype(np.float32)
+    y = (a[:, 0] + a[:, 1] > 7).astype(np.float32)
+    enc = OneHotEncoder(sparse_output=True).fit(from_numpy(a))
+    S = enc.transform(from_numpy(a))
+    assert isinstance(S, SparseDistArray) and S.device.type == "cuda"
+    lr = LogisticRegression(C=1.0, max_iter=60).fit(S, from_numpy(y))
+    acc = float((lr.predict(S).gather().cpu().numpy() == y).mean())
+    assert acc > 0.9
+    # sparse fit == dense fit on device
+    lrd = LogisticRegression(C=1.0, max_iter=60).fit(S.todense(),
+                                                     from_numpy(y))
+    np.testing.assert_allclose(lr.coef_.cpu().numpy(),
+                               lrd.coef_.cpu().numpy(), rtol=1e-3, atol=1e-3)
