@@ -52,13 +52,21 @@ def _is_pipeline(est) -> bool:
 
 
 class _SearchBase(BaseEstimator):
+    #: row*col threshold under which "auto" switches to candidate-parallel
+    _SMALL_DATA_CELLS = 2_000_000
+
     def __init__(self, estimator, scoring=None, cv=None, refit=True,
-                 return_train_score=False):
+                 return_train_score=False, candidate_parallel="auto"):
         self.estimator = estimator
         self.scoring = scoring
         self.cv = cv
         self.refit = refit
         self.return_train_score = return_train_score
+        # "auto": small data + world>1 -> one candidate subset per rank
+        # (each rank fits its candidates on the locally-gathered data, the
+        # reference's cluster-wide candidate fan-out); "ranks" forces it;
+        # "none" keeps every candidate a distributed fit.
+        self.candidate_parallel = candidate_parallel
 
     def _candidates(self) -> List[Dict]:
         raise NotImplementedError
@@ -101,13 +109,18 @@ class _SearchBase(BaseEstimator):
             est.fit(Xtr)
         return self._score(est, Xte, yte)
 
-    def fit(self, X, y=None, **fit_params):
-        X = check_array(X, dtype=None)
-        if y is not None:
-            y = check_array(y, dtype=None)
-        cv = check_cv(self.cv)
-        candidates = self._candidates()
-        n_splits = cv.get_n_splits(X, y)
+    def _use_candidate_parallel(self, X, n_candidates: int) -> bool:
+        if X.comm.world_size == 1 or n_candidates < 2:
+            return False
+        if self.candidate_parallel == "ranks":
+            return True
+        if self.candidate_parallel == "auto":
+            return (X.global_rows * max(1, X.ncols)
+                    <= self._SMALL_DATA_CELLS)
+        return False
+
+    def _score_grid(self, X, y, candidates, cv, n_splits) -> np.ndarray:
+        """scores[c, f] with every candidate fitted by ALL ranks."""
         scores = np.zeros((len(candidates), n_splits))
         for f, (tr_mask, te_mask) in enumerate(cv.split(X, y)):
             arrays = [X] + ([y] if y is not None else [])
@@ -120,6 +133,51 @@ class _SearchBase(BaseEstimator):
             for c, params in enumerate(candidates):
                 scores[c, f] = self._fit_candidate(
                     params, Xtr, ytr, Xte, yte, cache, f)
+        return scores
+
+    def _score_grid_rank_parallel(self, X, y, candidates, cv, n_splits
+                                  ) -> np.ndarray:
+        """Candidates fan out round-robin across ranks; each rank fits its
+        subset on the locally-gathered (small) data with a single-process
+        communicator, then scores are all-gathered (the reference's
+        cluster-wide candidate parallelism for sub-GPU-sized problems)."""
+        from ..parallel import comm as comm_mod
+        comm = X.comm
+        lcomm = comm_mod.local()
+        Xl = DistArray(X.gather(), X.global_rows, 0, lcomm)
+        yl = (DistArray(y.gather(), y.global_rows, 0, lcomm)
+              if y is not None else None)
+        mine = list(range(comm.rank, len(candidates), comm.world_size))
+        scores = np.full((len(candidates), n_splits), np.nan)
+        for f, (tr_mask, te_mask) in enumerate(cv.split(Xl, yl)):
+            arrays = [Xl] + ([yl] if yl is not None else [])
+            tr = mask_take(arrays, tr_mask)
+            te = mask_take(arrays, te_mask)
+            Xtr, Xte = tr[0], te[0]
+            ytr = tr[1] if yl is not None else None
+            yte = te[1] if yl is not None else None
+            cache: Dict[str, Any] = {}
+            for c in mine:
+                scores[c, f] = self._fit_candidate(
+                    candidates[c], Xtr, ytr, Xte, yte, cache, f)
+        parts = comm.all_gather_object({c: scores[c].tolist() for c in mine})
+        for part in parts:
+            for c, row in part.items():
+                scores[c] = row
+        return scores
+
+    def fit(self, X, y=None, **fit_params):
+        X = check_array(X, dtype=None)
+        if y is not None:
+            y = check_array(y, dtype=None)
+        cv = check_cv(self.cv)
+        candidates = self._candidates()
+        n_splits = cv.get_n_splits(X, y)
+        if self._use_candidate_parallel(X, len(candidates)):
+            scores = self._score_grid_rank_parallel(X, y, candidates, cv,
+                                                    n_splits)
+        else:
+            scores = self._score_grid(X, y, candidates, cv, n_splits)
         mean = scores.mean(axis=1)
         std = scores.std(axis=1)
         best = int(np.argmax(mean))
@@ -161,8 +219,10 @@ class _SearchBase(BaseEstimator):
 
 class GridSearchCV(_SearchBase):
     def __init__(self, estimator, param_grid, scoring=None, cv=None,
-                 refit=True, return_train_score=False):
-        super().__init__(estimator, scoring, cv, refit, return_train_score)
+                 refit=True, return_train_score=False,
+                 candidate_parallel="auto"):
+        super().__init__(estimator, scoring, cv, refit, return_train_score,
+                         candidate_parallel)
         self.param_grid = param_grid
 
     def _candidates(self):
@@ -172,8 +232,9 @@ class GridSearchCV(_SearchBase):
 class RandomizedSearchCV(_SearchBase):
     def __init__(self, estimator, param_distributions, n_iter=10,
                  scoring=None, cv=None, refit=True, random_state=0,
-                 return_train_score=False):
-        super().__init__(estimator, scoring, cv, refit, return_train_score)
+                 return_train_score=False, candidate_parallel="auto"):
+        super().__init__(estimator, scoring, cv, refit, return_train_score,
+                         candidate_parallel)
         self.param_distributions = param_distributions
         self.n_iter = n_iter
         self.random_state = random_state

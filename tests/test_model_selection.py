@@ -160,3 +160,30 @@ def test_hyperband_world2(mp_runner):
     X, y = _cls(n=1600, seed=8)
     mp_runner(_hyperband_world2, world=2,
               args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1]))
+
+
+def _grid_rank_parallel(rank, world, a_bytes, y_bytes, n, f):
+    X = np.frombuffer(a_bytes, dtype=np.float32).reshape(n, f).copy()
+    y = np.frombuffer(y_bytes, dtype=np.float32).copy()
+    gs = GridSearchCV(LogisticRegression(solver="lbfgs", max_iter=50),
+                      {"C": [0.01, 0.1, 1.0, 10.0]}, cv=3,
+                      candidate_parallel="ranks")
+    gs.fit(from_numpy(X), from_numpy(y))
+    # same data via the all-ranks path must give identical scores
+    gs2 = GridSearchCV(LogisticRegression(solver="lbfgs", max_iter=50),
+                       {"C": [0.01, 0.1, 1.0, 10.0]}, cv=3,
+                       candidate_parallel="none")
+    gs2.fit(from_numpy(X), from_numpy(y))
+    assert gs.best_params_ == gs2.best_params_
+    np.testing.assert_allclose(gs.cv_results_["mean_test_score"],
+                               gs2.cv_results_["mean_test_score"], atol=0.03)
+    # every candidate got a real score on every fold
+    assert not np.isnan(gs.cv_results_["mean_test_score"]).any()
+    pred = gs.predict(from_numpy(X)).gather().cpu().numpy()
+    assert pred.shape == (len(y),)
+
+
+def test_grid_search_candidate_parallel_world2(mp_runner):
+    X, y = _cls(1200, 5, seed=9)
+    mp_runner(_grid_rank_parallel, world=2,
+              args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1]))
