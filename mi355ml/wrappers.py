@@ -144,6 +144,12 @@ class Incremental(ParallelPostFit):
                 if _is_partitioned(X):
                     import scipy.sparse as sp
                     parts = X.local
+                    if y is not None and _is_partitioned(y) and \
+                            len(y.local) != len(parts):
+                        raise ValueError(
+                            f"X has {len(parts)} local partitions but y has "
+                            f"{len(y.local)}; X and y must be partitioned "
+                            "identically (same DistList split)")
                     xb = (sp.vstack(parts) if parts and
                           sp.issparse(parts[0]) else
                           (np.concatenate(parts) if parts else
@@ -154,6 +160,10 @@ class Incremental(ParallelPostFit):
                     xb = _to_numpy(X.local if X.ndim == 2
                                    else X.local[:, None])
                     yb = _to_numpy(y.local) if y is not None else None
+                if yb is not None and xb.shape[0] != len(yb):
+                    raise ValueError(
+                        f"local X block has {xb.shape[0]} rows but y has "
+                        f"{len(yb)}; X and y must shard identically")
                 if xb.shape[0]:
                     if classes is not None and yb is not None:
                         # sklearn accepts (identical) classes on every call

@@ -187,3 +187,21 @@ def test_grid_search_candidate_parallel_world2(mp_runner):
     X, y = _cls(1200, 5, seed=9)
     mp_runner(_grid_rank_parallel, world=2,
               args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1]))
+
+
+def test_incremental_mismatched_partitions_raises():
+    from mi355ml.core.bag import DistList
+    X = DistList([np.zeros((10, 3), dtype=np.float32),
+                  np.zeros((5, 3), dtype=np.float32)])
+    y = DistList([np.zeros(10, dtype=np.float32)])
+    inc = Incremental(SGDClassifier())
+    with pytest.raises(ValueError, match="partition"):
+        inc.fit(X, y, classes=[0.0, 1.0])
+
+
+def test_incremental_mismatched_rows_raises():
+    X = from_numpy(np.zeros((20, 3), dtype=np.float32))
+    y = from_numpy(np.zeros(12, dtype=np.float32))
+    inc = Incremental(SGDClassifier())
+    with pytest.raises(ValueError, match="shard"):
+        inc.fit(X, y, classes=[0.0, 1.0])
