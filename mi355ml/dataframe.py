@@ -85,6 +85,75 @@ def groupby_agg(frames: DistList, by: Union[str, List[str]],
     return _combine(agg, agg)  # min of mins / max of maxes
 
 
+def _key_frame(part: pd.DataFrame, by) -> pd.Series:
+    cols = [by] if isinstance(by, str) else list(by)
+    # deterministic cross-process hash (fixed-key siphash)
+    return pd.util.hash_pandas_object(part[cols], index=False)
+
+
+def shuffle(frames: DistList, on: Union[str, List[str]]) -> DistList:
+    """Hash-partition rows by key so all rows of a key land on ONE rank
+    (the reference's rearrange_by_column task shuffle,
+    dask/dataframe/shuffle.py:443). Enables exact groupby-apply and joins.
+
+    Dataframe partitions are control-plane data here (the dense data plane
+    is DistArray), so the exchange is one object all-gather; each rank
+    keeps the slices hashed to it."""
+    comm = frames.comm
+    W = comm.world_size
+    if W == 1:
+        parts = [p for p in frames.local if p is not None and len(p)]
+        # one partition so every key group is contiguous for groupby-apply
+        return DistList([pd.concat(parts) if parts else pd.DataFrame()],
+                        comm)
+    outgoing: List[List[pd.DataFrame]] = [[] for _ in range(W)]
+    for part in frames.local:
+        if part is None or not len(part):
+            continue
+        dest = (_key_frame(part, on) % W).to_numpy()
+        for r in range(W):
+            sel = part[dest == r]
+            if len(sel):
+                outgoing[r].append(sel)
+    template = next((p.iloc[:0] for p in frames.local
+                     if p is not None and p.shape[1]), None)
+    gathered = comm.all_gather_object(
+        ([pd.concat(o) if o else None for o in outgoing], template))
+    mine = [out[comm.rank] for out, _ in gathered
+            if out[comm.rank] is not None]
+    if mine:
+        return DistList([pd.concat(mine)], comm)
+    empty = next((t for _, t in gathered if t is not None), pd.DataFrame())
+    return DistList([empty], comm)
+
+
+def groupby_apply(frames: DistList, by: Union[str, List[str]],
+                  fn: Callable) -> DistList:
+    """Exact groupby-apply: shuffle by key, then pandas groupby per
+    partition (every group is wholly on one rank after the shuffle)."""
+    shuffled = shuffle(frames, by)
+
+    def _apply(part: pd.DataFrame):
+        if part is None or not len(part):
+            return part
+        return part.groupby(by, group_keys=False).apply(fn)
+
+    return shuffled.map(_apply)
+
+
+def merge(left: DistList, right: DistList, on: Union[str, List[str]],
+          how: str = "inner") -> DistList:
+    """Distributed join: co-shuffle both frames by key, then per-rank
+    pandas merge (dask.dataframe.multi hash-join analog)."""
+    ls = shuffle(left, on)
+    rs = shuffle(right, on)
+    lp = ls.local[0] if ls.local else pd.DataFrame()
+    rp = rs.local[0] if rs.local else pd.DataFrame()
+    if not len(lp) and not len(rp):
+        return DistList([pd.DataFrame()], left.comm)
+    return DistList([lp.merge(rp, on=on, how=how)], left.comm)
+
+
 def value_counts(frames: DistList, column: str) -> pd.Series:
     def _vc(part):
         return part[column].value_counts() if len(part) else None

@@ -115,3 +115,68 @@ def _df_world2(rank, world):
 
 def test_dataframe_world2(mp_runner):
     mp_runner(_df_world2, world=2)
+
+
+def _toy_frames(n=500, seed=0, nparts=3):
+    from mi355ml.dataframe import from_pandas
+    rng = np.random.default_rng(seed)
+    df = pd.DataFrame({
+        "k": rng.integers(0, 17, n),
+        "v": rng.standard_normal(n),
+        "w": rng.integers(0, 5, n),
+    })
+    return df, from_pandas(df, npartitions=nparts)
+
+
+def test_dataframe_shuffle_groupby_apply():
+    from mi355ml.dataframe import concat, groupby_apply, shuffle
+    df, frames = _toy_frames()
+    sh = shuffle(frames, "k")
+    got = concat(sh).sort_values(["k", "v"]).reset_index(drop=True)
+    exp = df.sort_values(["k", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp)
+    # groupby-apply: demean within each key group (needs co-located groups)
+    out = groupby_apply(frames, "k", lambda g: g.assign(v=g.v - g.v.mean()))
+    got = concat(out)
+    exp2 = df.groupby("k", group_keys=False).apply(
+        lambda g: g.assign(v=g.v - g.v.mean()))
+    got = got.sort_values(["k", "v"]).reset_index(drop=True)
+    exp2 = exp2.sort_values(["k", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp2)
+
+
+def test_dataframe_merge():
+    from mi355ml.dataframe import concat, from_pandas, merge
+    df, frames = _toy_frames(300, seed=1)
+    rng = np.random.default_rng(2)
+    other = pd.DataFrame({"k": np.arange(17), "name": rng.integers(0, 9, 17)})
+    oframes = from_pandas(other, npartitions=2)
+    for how in ("inner", "left", "outer"):
+        got = concat(merge(frames, oframes, on="k", how=how))
+        exp = df.merge(other, on="k", how=how)
+        got = got.sort_values(["k", "v"]).reset_index(drop=True)
+        exp = exp.sort_values(["k", "v"]).reset_index(drop=True)
+        pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _shuffle_world2(rank, world, n):
+    from mi355ml.dataframe import concat, groupby_agg, merge, shuffle
+    from mi355ml.dataframe import from_pandas
+    rng = np.random.default_rng(5)
+    df = pd.DataFrame({"k": rng.integers(0, 11, n),
+                       "v": rng.standard_normal(n)})
+    frames = from_pandas(df, npartitions=4)
+    sh = shuffle(frames, "k")
+    # disjoint key ownership across ranks
+    local_keys = set()
+    for p in sh.local:
+        local_keys |= set(p["k"].unique().tolist())
+    others = sh.comm.all_gather_object(local_keys)
+    assert local_keys.isdisjoint(others[1 - rank])
+    got = concat(sh).sort_values(["k", "v"]).reset_index(drop=True)
+    exp = df.sort_values(["k", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp)
+
+
+def test_dataframe_shuffle_world2(mp_runner):
+    mp_runner(_shuffle_world2, world=2, args=(400,))
