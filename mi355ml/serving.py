@@ -33,7 +33,8 @@ def _to_plain(res) -> List:
     return np.asarray(res).tolist()
 
 
-def create_app(model: Any, max_batch: int = 65536):
+def create_app(model: Any, max_batch: int = 65536,
+               token: Optional[str] = None):
     from starlette.applications import Starlette
     from starlette.responses import JSONResponse, PlainTextResponse
     from starlette.routing import Route
@@ -63,6 +64,11 @@ def create_app(model: Any, max_batch: int = 65536):
 
     def _method(name: str):
         async def endpoint(request):
+            if token is not None:
+                auth = request.headers.get("authorization", "")
+                if auth != f"Bearer {token}":
+                    return JSONResponse({"detail": "unauthorized"},
+                                        status_code=401)
             if not hasattr(model, name):
                 return JSONResponse({"detail": f"model has no {name}()"},
                                     status_code=400)
@@ -117,9 +123,13 @@ def main(argv: Optional[list] = None):
     ap.add_argument("--port", type=int, default=8000)
     ap.add_argument("--max-batch", type=int, default=65536,
                     help="largest accepted request batch (rows)")
+    ap.add_argument("--token", default=None,
+                    help="require 'Authorization: Bearer <token>' on "
+                         "inference endpoints")
     args = ap.parse_args(argv)
     model = joblib.load(args.model)
-    uvicorn.run(create_app(model, max_batch=args.max_batch),
+    uvicorn.run(create_app(model, max_batch=args.max_batch,
+                           token=args.token),
                 host=args.host, port=args.port)
 
 

@@ -47,3 +47,18 @@ def test_serving_batch_limit():
     assert r.status_code == 413
     r = client.post("/predict", json={"data": X[:50].tolist()})
     assert r.status_code == 200
+
+
+def test_serving_auth():
+    rng = np.random.default_rng(2)
+    X = rng.standard_normal((100, 3)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    lr = LogisticRegression(max_iter=10).fit(from_numpy(X), from_numpy(y))
+    client = TestClient(create_app(lr, token="s3cret"))
+    r = client.post("/predict", json={"data": X[:5].tolist()})
+    assert r.status_code == 401
+    r = client.post("/predict", json={"data": X[:5].tolist()},
+                    headers={"Authorization": "Bearer s3cret"})
+    assert r.status_code == 200
+    # health stays open
+    assert client.get("/health").status_code == 200
