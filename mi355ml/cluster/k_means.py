@@ -86,14 +86,23 @@ def k_init(X: DistArray, n_clusters: int, random_state: int = 0,
     centers = _global_row(X, first)[None, :]
 
     def _d2_against(cset: torch.Tensor):
-        """Per-local-row squared distance to the closest of cset."""
-        outs = []
+        """Per-local-row (argmin-in-cset, squared distance to closest)."""
+        labs, outs = [], []
         for _, blk in _iter_blocks(X):
-            _, d2 = closest_center(blk, cset.to(blk.device))
+            lab, d2 = closest_center(blk, cset.to(blk.device))
+            labs.append(lab.to(dev))
             outs.append(d2.to(dev))
-        return torch.cat(outs) if len(outs) > 1 else outs[0]
+        if len(outs) > 1:
+            return torch.cat(labs), torch.cat(outs)
+        return labs[0], outs[0]
 
-    d2 = _d2_against(centers).double()
+    _, d2 = _d2_against(centers)
+    d2 = d2.double()
+    # running nearest-candidate label per local row: maintained across
+    # rounds so the final weighting pass against the full grown candidate
+    # set (the most expensive single pass — candidates ≫ the fused
+    # kernel's k limit) is never needed.
+    best_lab = torch.zeros(d2.shape[0], dtype=torch.int64, device=dev)
     phi_t = d2.sum().reshape(1)
     if comm.world_size > 1:
         comm.all_reduce(phi_t)
@@ -119,16 +128,21 @@ def k_init(X: DistArray, n_clusters: int, random_state: int = 0,
         parts = comm.all_gather_varlen(cand.contiguous())
         new = torch.cat(parts, dim=0) if comm.world_size > 1 else cand
         if new.shape[0]:
+            offset = centers.shape[0]
             centers = torch.cat([centers, new.to(centers.dtype)], dim=0)
-            d2 = torch.minimum(d2, _d2_against(new).double())
+            lab_new, d2_new = _d2_against(new)
+            d2_new = d2_new.double()
+            upd = d2_new < d2
+            best_lab = torch.where(upd, lab_new.long() + offset, best_lab)
+            d2 = torch.minimum(d2, d2_new)
             phi_t = d2.sum().reshape(1)
             if comm.world_size > 1:
                 comm.all_reduce(phi_t)
             phi = float(phi_t[0])
 
-    # weight candidates by assignment counts, then local weighted KMeans
-    labels = _shard_labels(X, centers)
-    w = torch.bincount(labels.long(),
+    # weight candidates by assignment counts (tracked incrementally above),
+    # then local weighted KMeans
+    w = torch.bincount(best_lab,
                        minlength=centers.shape[0]).double().to(dev)
     if comm.world_size > 1:
         comm.all_reduce(w)
