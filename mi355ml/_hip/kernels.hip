@@ -179,9 +179,18 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
   float* xtbuf = sw + ((d + 3) & ~3);              // nbuf x [TR * ld]
   float* ytbuf = xtbuf + nbuf * GLM_TR * ld;       // nbuf x [TR]
   float* rres = ytbuf + nbuf * GLM_TR;             // [TR]
-  double* red = (double*)(smem_raw +
-      (((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
-        sizeof(float) + 15) & ~15));
+  // red[NT] lives INSIDE xtbuf (past the gred window) when it fits: both
+  // are only touched after the tile loop, and the 2 KB saved is the
+  // 2->3 WG/CU occupancy step at d~100. Tiny-d tiles fall back to a
+  // dedicated allocation (launcher mirrors this condition exactly).
+  const bool red_in_xtbuf =
+      (size_t)nbuf * GLM_TR * ld * sizeof(float) >=
+      (4 * (size_t)d + NT) * sizeof(double);
+  double* red = red_in_xtbuf
+      ? (double*)xtbuf + 4 * d
+      : (double*)(smem_raw +
+            (((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
+              sizeof(float) + 15) & ~15));
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -756,8 +765,12 @@ void launch_glm_value_grad(const float* x, const float* y, const float* w,
   // otherwise odd stride (conflict-free, scalar staging).
   const int ld = ((d & 3) == 0) ? d : (d | 1);
   auto smem_for = [&](int nbuf) {
+    const bool red_in_xtbuf =
+        (size_t)nbuf * GLM_TR * ld * sizeof(float) >=
+        (4 * (size_t)d + NT) * sizeof(double);
     return ((((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
-              sizeof(float) + 15) & ~15)) + NT * sizeof(double);
+              sizeof(float) + 15) & ~15)) +
+           (red_in_xtbuf ? 0 : NT * sizeof(double));
   };
   const int nbuf = (smem_for(2) <= 158 * 1024) ? 2 : 1;
   const size_t smem = smem_for(nbuf);
@@ -792,9 +805,12 @@ void launch_glm_probe(const float* x, const float* y, const float* w,
                       float b, long m, int d, int phases, double* slab,
                       int nblocks, hipStream_t stream) {
   const int ld = ((d & 3) == 0) ? d : (d | 1);
+  const bool red_in_xtbuf =
+      (size_t)2 * GLM_TR * ld * sizeof(float) >=
+      (4 * (size_t)d + NT) * sizeof(double);
   const size_t smem =
       ((((((d + 3) & ~3) + 2 * GLM_TR * ld + 3 * GLM_TR) * sizeof(float) +
-         15) & ~15)) + NT * sizeof(double);
+         15) & ~15)) + (red_in_xtbuf ? 0 : NT * sizeof(double));
 #define GP(PH)                                                              \
   case PH:                                                                  \
     hipLaunchKernelGGL((glm_value_grad_k<0, PH>), dim3(nblocks), dim3(NT),  \
