@@ -206,11 +206,19 @@ def col_percentiles_tdigest(X: DistArray, qs: Sequence[float],
     from ..core.spill import SpilledDistArray
     dev = X.device
     F = X.shape[1] if len(X.shape) > 1 else 1
-    if isinstance(X, SpilledDistArray):
-        blocks = (b for _, b in X.device_blocks())
-    else:
-        blocks = iter([X._as2d()])
-    means, wsum, mins, maxs = _tdigest_shard(blocks, F, dev, delta)
+    MAXROWS = 4 << 20  # bound the f64 sort workspace per chunk
+
+    def _chunks():
+        if isinstance(X, SpilledDistArray):
+            src = (b for _, b in X.device_blocks())
+        else:
+            src = iter([X._as2d()])
+        for b in src:
+            b2 = b if b.dim() == 2 else b[:, None]
+            for lo in range(0, max(b2.shape[0], 1), MAXROWS):
+                yield b2[lo:lo + MAXROWS]
+
+    means, wsum, mins, maxs = _tdigest_shard(_chunks(), F, dev, delta)
     if X.comm.world_size > 1:
         g_m = torch.cat(X.comm.all_gather(means.contiguous()), dim=1)
         g_w = torch.cat(X.comm.all_gather(wsum.contiguous()), dim=1)
@@ -241,15 +249,23 @@ def col_percentiles(X: DistArray, qs: Sequence[float],
     q_ext_t = torch.tensor(q_ext, dtype=torch.float64)
     local_curves = []
     local_ns = []
+    # torch.quantile rejects slices over 2^24 elements: large shards
+    # contribute one curve per row block (the merge handles any number of
+    # curves — same approximation class as more chunks in the reference)
+    MAXROWS = 1 << 24
     if isinstance(X, SpilledDistArray):
         for _, blk in X.device_blocks():
             b2 = blk if blk.dim() == 2 else blk[:, None]
-            local_curves.append(_shard_curve(b2, q_ext_t))
-            local_ns.append(float(b2.shape[0]))
+            for lo in range(0, max(b2.shape[0], 1), MAXROWS):
+                c = b2[lo:lo + MAXROWS]
+                local_curves.append(_shard_curve(c, q_ext_t))
+                local_ns.append(float(c.shape[0]))
     else:
         x = X._as2d()
-        local_curves.append(_shard_curve(x, q_ext_t))
-        local_ns.append(float(x.shape[0]))
+        for lo in range(0, max(x.shape[0], 1), MAXROWS):
+            c = x[lo:lo + MAXROWS]
+            local_curves.append(_shard_curve(c, q_ext_t))
+            local_ns.append(float(c.shape[0]))
     # pad to a fixed per-rank curve count for the dense all-gather
     nloc = torch.tensor([len(local_curves)], dtype=torch.int64, device=dev)
     if X.comm.world_size > 1:
