@@ -93,11 +93,29 @@ class SparseDistArray(DistArray):
 
     # ------------------------------------------------------------ compute
     def _row_ids(self) -> torch.Tensor:
-        """[nnz] row index of every stored value (expanded crow)."""
-        m = self.local_rows
-        lengths = self.crow[1:] - self.crow[:-1]
-        return torch.repeat_interleave(
-            torch.arange(m, device=self.crow.device), lengths)
+        """[nnz] row index of every stored value (expanded crow; cached —
+        the CSR structure is immutable)."""
+        if getattr(self, "_row_ids_c", None) is None:
+            m = self.local_rows
+            lengths = self.crow[1:] - self.crow[:-1]
+            self._row_ids_c = torch.repeat_interleave(
+                torch.arange(m, device=self.crow.device), lengths)
+        return self._row_ids_c
+
+    def _col_groups(self):
+        """Cached column-ordered (CSC-like) view for Xᵀ reductions:
+        (vals_sorted, rows_sorted, uniq_cols, offsets)."""
+        if getattr(self, "_col_groups_c", None) is None:
+            order = torch.argsort(self.col, stable=True)
+            colz = self.col[order]
+            uniq, counts = torch.unique_consecutive(colz,
+                                                    return_counts=True)
+            offsets = torch.zeros(uniq.shape[0] + 1, dtype=torch.int64,
+                                  device=self.device)
+            offsets[1:] = counts.cumsum(0)
+            self._col_groups_c = (self.vals[order], self._row_ids()[order],
+                                  uniq, offsets)
+        return self._col_groups_c
 
     def dot(self, w: torch.Tensor) -> DistArray:
         """X·w (replicated w) — gather-multiply + ordered per-row sum.
@@ -125,17 +143,13 @@ class SparseDistArray(DistArray):
         """Per-shard Xᵀ·v partial (no collective — solvers fuse their own)."""
         r = v.local if isinstance(v, DistArray) else v
         r = r.to(device=self.device, dtype=dtype)
-        contrib = self.vals.to(dtype) * r[self._row_ids()]
-        # column-sorted ordered reduction (deterministic on GPU too)
-        order = torch.argsort(self.col, stable=True)
-        colz = self.col[order]
+        # column-sorted ordered reduction (deterministic on GPU too); the
+        # CSC-like permuted view of the static structure is cached
         g = torch.zeros(self._ncols, dtype=dtype, device=self.device)
-        if colz.numel():
-            uniq, counts = torch.unique_consecutive(colz, return_counts=True)
-            offsets = torch.zeros(uniq.shape[0] + 1, dtype=torch.int64,
-                                  device=self.device)
-            offsets[1:] = counts.cumsum(0)
-            sums = torch.segment_reduce(contrib[order], "sum",
+        if self.col.numel():
+            vals_s, rows_s, uniq, offsets = self._col_groups()
+            contrib = vals_s.to(dtype) * r[rows_s]
+            sums = torch.segment_reduce(contrib, "sum",
                                         offsets=offsets, axis=0, initial=0)
             g[uniq] = sums
         return g

@@ -108,19 +108,21 @@ def _tdigest_build(vals: torch.Tensor, weights: torch.Tensor, delta: int
 
     vals/weights: [N, F] float64 (weight 0 = missing; their value must sort
     last, i.e. be +inf). Returns fixed-width digests; empty buckets have
-    weight 0.
+    weight 0. Sorts in [F, N] layout (contiguous per-column rows — a
+    dim-0 strided sort of the [N, F] image is ~50x slower).
     """
     N, F = vals.shape
     B = delta // 2 + 1
-    v, order = torch.sort(vals, dim=0)
-    w = torch.gather(weights, 0, order)
-    cum = w.cumsum(dim=0)
-    total = cum[-1:].clamp(min=1e-300)
+    vt = vals.t().contiguous()
+    v, order = torch.sort(vt, dim=1)
+    w = torch.gather(weights.t().contiguous(), 1, order)
+    cum = w.cumsum(dim=1)
+    total = cum[:, -1:].clamp(min=1e-300)
     q_mid = (cum - 0.5 * w) / total
     k = (delta / (2.0 * torch.pi)) * torch.asin(
         (2.0 * q_mid - 1.0).clamp(-1.0, 1.0)) + delta / 4.0
     bucket = k.floor().long().clamp(0, B - 1)
-    flat = bucket + torch.arange(F, device=vals.device)[None, :] * B
+    flat = bucket + torch.arange(F, device=vals.device)[:, None] * B
     wsum = torch.zeros(F * B, dtype=torch.float64, device=vals.device)
     wvsum = torch.zeros_like(wsum)
     wsum.scatter_add_(0, flat.reshape(-1), w.reshape(-1))
