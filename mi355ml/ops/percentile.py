@@ -122,13 +122,24 @@ def _tdigest_build(vals: torch.Tensor, weights: torch.Tensor, delta: int
     k = (delta / (2.0 * torch.pi)) * torch.asin(
         (2.0 * q_mid - 1.0).clamp(-1.0, 1.0)) + delta / 4.0
     bucket = k.floor().long().clamp(0, B - 1)
-    flat = bucket + torch.arange(F, device=vals.device)[:, None] * B
+    flat = (bucket +
+            torch.arange(F, device=vals.device)[:, None] * B).reshape(-1)
+    # flat is globally non-decreasing (buckets rise within a row, rows are
+    # offset by B): reduce contiguous runs instead of scatter_add — the
+    # N-into-F*B atomic contention was 100-1000x slower, and this is
+    # deterministic (ordered segment sums)
+    uniqb, counts = torch.unique_consecutive(flat, return_counts=True)
+    offsets = torch.zeros(uniqb.shape[0] + 1, dtype=torch.int64,
+                          device=vals.device)
+    offsets[1:] = counts.cumsum(0)
     wsum = torch.zeros(F * B, dtype=torch.float64, device=vals.device)
     wvsum = torch.zeros_like(wsum)
-    wsum.scatter_add_(0, flat.reshape(-1), w.reshape(-1))
-    wvsum.scatter_add_(0, flat.reshape(-1),
-                       (w * torch.where(torch.isfinite(v), v,
-                                        torch.zeros_like(v))).reshape(-1))
+    wsum[uniqb] = torch.segment_reduce(w.reshape(-1), "sum",
+                                       offsets=offsets, axis=0, initial=0)
+    wvf = (w * torch.where(torch.isfinite(v), v,
+                           torch.zeros_like(v))).reshape(-1)
+    wvsum[uniqb] = torch.segment_reduce(wvf, "sum", offsets=offsets,
+                                        axis=0, initial=0)
     wsum = wsum.reshape(F, B)
     means = wvsum.reshape(F, B) / wsum.clamp(min=1e-300)
     return means, wsum
