@@ -147,6 +147,30 @@ class _StreamHandle:
         return self.tensor
 
 
+def _elastic_store(rank: int, world: int, restart: int,
+                   timeout) -> "dist.PrefixStore":
+    """TCPStore + per-attempt prefix for torchelastic respawns. The master
+    retries the bind: the previous attempt's rank-0 process may hold the
+    port for a short window while it dies."""
+    import time as _time
+    host = os.environ.get("MASTER_ADDR", "127.0.0.1")
+    port = int(os.environ.get("MASTER_PORT", "29531"))
+    deadline = _time.monotonic() + timeout.total_seconds()
+    last = None
+    while True:
+        try:
+            ts = dist.TCPStore(host, port, world, rank == 0,
+                               timeout=timeout)
+            break
+        except Exception as e:  # bind/connect race with the dying attempt
+            last = e
+            if _time.monotonic() > deadline:
+                raise RuntimeError(
+                    f"elastic store setup failed after retries: {last}")
+            _time.sleep(0.25)
+    return dist.PrefixStore(f"mi355ml/attempt-{restart}", ts)
+
+
 def _pick_device(local_rank: int) -> torch.device:
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
@@ -175,10 +199,22 @@ def init(backend: Optional[str] = None) -> Communicator:
                 RuntimeWarning, stacklevel=2)
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29531")
-        dist.init_process_group(
-            backend=backend, rank=rank, world_size=world,
-            timeout=datetime.timedelta(seconds=float(config.get("comm.timeout"))),
-        )
+        timeout = datetime.timedelta(
+            seconds=float(config.get("comm.timeout")))
+        restart = os.environ.get("TORCHELASTIC_RESTART_COUNT")
+        if restart is not None:
+            # under torchrun --max-restarts, namespace the store per attempt:
+            # a respawned rank must never read the previous attempt's gloo
+            # listener addresses (a dying predecessor's store can serve
+            # stale keys for a short window -> connectFullMesh 'connection
+            # refused' flakes on restart)
+            store = _elastic_store(rank, world, int(restart), timeout)
+            dist.init_process_group(backend=backend, rank=rank,
+                                    world_size=world, store=store,
+                                    timeout=timeout)
+        else:
+            dist.init_process_group(backend=backend, rank=rank,
+                                    world_size=world, timeout=timeout)
     _DEFAULT = Communicator(rank, world, device)
     return _DEFAULT
 
