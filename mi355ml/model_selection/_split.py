@@ -31,10 +31,17 @@ def train_test_split(*arrays, test_size: float = None, train_size: float = None,
         test_size = 0.25 if train_size is None else 1.0 - train_size
     arrays = [check_array(a, dtype=None) for a in arrays]
     X0 = arrays[0]
-    g = mlrand.generator(int(random_state or 0), shard_id=X0.comm.rank,
-                         stream=13, device=X0.device)
-    u = torch.rand(X0.local.shape[0], generator=g, device=X0.device)
-    mask = u < test_size
+    if shuffle:
+        g = mlrand.generator(int(random_state or 0), shard_id=X0.comm.rank,
+                             stream=13, device=X0.device)
+        u = torch.rand(X0.local.shape[0], generator=g, device=X0.device)
+        mask = u < test_size
+    else:
+        # contiguous split (reference shuffle=False: trailing rows = test)
+        lo = X0.row_offset
+        idx = torch.arange(lo, lo + X0.local.shape[0], device=X0.device)
+        cut = int(round(X0.global_rows * (1.0 - float(test_size))))
+        mask = idx >= cut
     out = []
     for a in arrays:
         if a.local.shape[0] != X0.local.shape[0]:
@@ -82,6 +89,20 @@ class KFold:
         n = X.global_rows
         lo = X.row_offset
         m = X.local.shape[0]
+        if self.shuffle:
+            # rows keep their rank; fold membership is a deterministic
+            # per-row random draw (the no-row-movement analog of the
+            # reference's shuffled block folds; folds are equal-sized in
+            # expectation)
+            g = mlrand.generator(int(self.random_state or 0),
+                                 shard_id=X.comm.rank, stream=29,
+                                 device=X.device)
+            fold = torch.randint(0, self.n_splits, (m,), generator=g,
+                                 device=X.device)
+            for f in range(self.n_splits):
+                te = fold == f
+                yield ~te, te
+            return
         idx = torch.arange(lo, lo + m, device=X.device)
         base, rem = divmod(n, self.n_splits)
         start = 0

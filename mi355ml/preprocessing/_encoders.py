@@ -74,22 +74,31 @@ class OrdinalEncoder(_BaseEncoder):
 
 class OneHotEncoder(_BaseEncoder):
     def __init__(self, categories="auto", dtype=torch.float32,
-                 handle_unknown: str = "error", sparse_output: bool = False):
+                 handle_unknown: str = "error", sparse_output: bool = False,
+                 drop=None):
         super().__init__(categories, dtype, handle_unknown)
         self.sparse_output = sparse_output
+        self.drop = drop  # None or "first" (collinearity guard)
+
+    def _sizes(self):
+        d1 = 1 if self.drop == "first" else 0
+        return [int(c.shape[0]) - d1 for c in self.categories_]
 
     def transform(self, X):
         check_is_fitted(self, "categories_")
+        if self.drop not in (None, "first"):
+            raise ValueError("drop must be None or 'first'")
         X = check_array(X, dtype=None)
-        sizes = [int(c.shape[0]) for c in self.categories_]
+        sizes = self._sizes()
         offsets = np.concatenate([[0], np.cumsum(sizes)])
         total = int(offsets[-1])
         dt = self.dtype
         if self.sparse_output:
             return self._transform_sparse(X, offsets, total)
+        d1 = 1 if self.drop == "first" else 0
 
         def _oh(t):
-            codes = self._codes(t)
+            codes = self._codes(t) - d1  # dropped category -> code -1
             out = torch.zeros(t.shape[0], total, dtype=dt, device=t.device)
             for f in range(codes.shape[1]):
                 c = codes[:, f]
@@ -108,6 +117,8 @@ class OneHotEncoder(_BaseEncoder):
         m, F = t.shape
         dev = t.device
         codes = self._codes(t)                                   # [m, F]
+        if self.drop == "first":
+            codes = codes - 1
         off = torch.as_tensor(offsets[:-1], device=dev)
         valid = codes >= 0
         nnz_per_row = valid.sum(dim=1)
@@ -123,7 +134,8 @@ class OneHotEncoder(_BaseEncoder):
         names = (list(input_features) if input_features is not None
                  else [f"x{i}" for i in range(self.n_features_in_)])
         out = []
+        d1 = 1 if self.drop == "first" else 0
         for f, cats in enumerate(self.categories_):
-            for v in cats.cpu().tolist():
+            for v in cats.cpu().tolist()[d1:]:
                 out.append(f"{names[f]}_{v}")
         return np.asarray(out, dtype=object)

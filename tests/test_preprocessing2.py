@@ -197,3 +197,35 @@ def _tdigest_world2(rank, world, a_bytes, shape):
 def test_tdigest_world2(mp_runner):
     a = _data(30_000, 4, seed=7)
     mp_runner(_tdigest_world2, world=2, args=(a.tobytes(), a.shape))
+
+
+def test_onehot_drop_first():
+    rng = np.random.default_rng(8)
+    a = rng.integers(0, 5, (2000, 3)).astype(np.float32)
+    ours = OneHotEncoder(drop="first").fit(from_numpy(a))
+    ref = SkOneHot(drop="first", sparse_output=False).fit(a)
+    got = ours.transform(from_numpy(a)).gather().cpu().numpy()
+    np.testing.assert_array_equal(got, ref.transform(a))
+    assert len(ours.get_feature_names_out()) == got.shape[1]
+    # sparse drop path agrees with dense
+    sp_got = OneHotEncoder(drop="first", sparse_output=True).fit(
+        from_numpy(a)).transform(from_numpy(a)).gather().toarray()
+    np.testing.assert_array_equal(sp_got, got)
+
+
+def test_split_no_shuffle_and_kfold_shuffle():
+    from mi355ml.model_selection import train_test_split, KFold
+    a = np.arange(100, dtype=np.float32)[:, None]
+    Xtr, Xte = train_test_split(from_numpy(a), test_size=0.3, shuffle=False)
+    tr = Xtr.gather().cpu().numpy().ravel()
+    te = Xte.gather().cpu().numpy().ravel()
+    np.testing.assert_array_equal(tr, np.arange(70))
+    np.testing.assert_array_equal(te, np.arange(70, 100))
+    kf = KFold(n_splits=4, shuffle=True, random_state=0)
+    X = from_numpy(np.zeros((1000, 2), dtype=np.float32))
+    seen = np.zeros(1000)
+    for tr_m, te_m in kf.split(X):
+        te_np = te_m.cpu().numpy()
+        assert 150 < te_np.sum() < 350  # ~250 each in expectation
+        seen += te_np
+    np.testing.assert_array_equal(seen, 1.0)  # every row in exactly 1 fold
