@@ -45,11 +45,20 @@ class Communicator:
             self._comm_stream = torch.cuda.Stream(device=device)
 
     # -- collectives ------------------------------------------------------
+    def _count(self, kind: str, t: torch.Tensor) -> None:
+        """Per-collective call/byte counters (diagnostics.COUNTERS; shows
+        up in serving /metrics — SURVEY §5.5 observability)."""
+        from ..diagnostics import COUNTERS
+        COUNTERS.inc(f"comm.{kind}.calls")
+        COUNTERS.inc(f"comm.{kind}.bytes",
+                     float(t.numel() * t.element_size()))
+
     def all_reduce(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         if self.world_size == 1:
             return t
         ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
                "min": dist.ReduceOp.MIN}
+        self._count("all_reduce", t)
         dist.all_reduce(t, op=ops[op], group=self.group)
         return t
 
@@ -70,6 +79,7 @@ class Communicator:
     def all_gather(self, t: torch.Tensor) -> List[torch.Tensor]:
         if self.world_size == 1:
             return [t]
+        self._count("all_gather", t)
         out = [torch.empty_like(t) for _ in range(self.world_size)]
         dist.all_gather(out, t.contiguous(), group=self.group)
         return out
@@ -93,6 +103,7 @@ class Communicator:
     def broadcast(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.world_size == 1:
             return t
+        self._count("broadcast", t)
         dist.broadcast(t, src=src, group=self.group)
         return t
 

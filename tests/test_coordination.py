@@ -84,3 +84,21 @@ def _futures_body(rank, world):
 
 def test_futures_world2(mp_runner):
     mp_runner(_futures_body, world=2)
+
+
+def _comm_counters_world2(rank, world):
+    import torch
+    from mi355ml.diagnostics import COUNTERS
+    from mi355ml.parallel import comm as comm_mod
+    comm = comm_mod.default()
+    t = torch.ones(100, dtype=torch.float64)
+    comm.all_reduce(t)
+    comm.all_gather(t)
+    snap = COUNTERS.snapshot()
+    assert snap["comm.all_reduce.calls"] >= 1
+    assert snap["comm.all_reduce.bytes"] >= 800
+    assert snap["comm.all_gather.calls"] >= 1
+
+
+def test_comm_counters_world2(mp_runner):
+    mp_runner(_comm_counters_world2, world=2, args=())
