@@ -150,3 +150,20 @@ def test_sparse_onehot_logreg_gpu():
                                                      from_numpy(y))
     np.testing.assert_allclose(lr.coef_.cpu().numpy(),
                                lrd.coef_.cpu().numpy(), rtol=1e-3, atol=1e-3)
+
+
+def test_sparse_xtdot_bitwise_deterministic_gpu():
+    import scipy.sparse as sp
+    from mi355ml.core.sparse import from_scipy
+    rng = np.random.default_rng(13)
+    M = sp.random(2_000_000, 300, density=0.02, random_state=1,
+                  format="csr", dtype=np.float32)
+    S = from_scipy(M)
+    r = torch.randn(2_000_000, device="cuda", generator=torch.Generator(
+        device="cuda").manual_seed(0))
+    g1 = S.xt_dot(r).cpu().numpy()
+    g2 = S.xt_dot(r).cpu().numpy()
+    assert (g1 == g2).all()  # ordered segment sums, no atomics
+    z1 = S.dot(torch.ones(300, device="cuda")).gather().cpu().numpy()
+    z2 = S.dot(torch.ones(300, device="cuda")).gather().cpu().numpy()
+    assert (z1 == z2).all()
