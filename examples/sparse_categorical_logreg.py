@@ -1,0 +1,42 @@
+"""Categorical features at scale without densifying: OneHotEncoder with
+sparse CSR output feeding the GLM solvers directly.
+
+The one-hot image of 2e7 rows x 8 categorical columns (64 levels each) is
+41 GB dense but 1.9 GB as CSR shards; the lbfgs fit consumes the CSR
+structure through deterministic segment-reduce SpMV (core/sparse.py) and
+never materializes the dense matrix in HBM.
+
+Run on 1 GPU:    python examples/sparse_categorical_logreg.py
+Run on 8 GPUs:   python -m mi355ml.parallel.launch --gpus 8 \
+                     examples/sparse_categorical_logreg.py
+"""
+
+import numpy as np
+
+from mi355ml.core.array import from_numpy
+from mi355ml.linear_model import LogisticRegression
+from mi355ml.parallel import comm as comm_mod
+from mi355ml.preprocessing import OneHotEncoder
+
+comm = comm_mod.init()
+
+rng = np.random.default_rng(0)
+n = 2_000_000  # per-process demo size; scale freely
+cats = rng.integers(0, 64, (n, 8)).astype(np.float32)
+y = (cats[:, 0] + cats[:, 1] > 62).astype(np.float32)
+
+X = from_numpy(cats)
+enc = OneHotEncoder(sparse_output=True).fit(X)
+S = enc.transform(X)  # SparseDistArray: CSR shards, one per rank
+if comm.rank == 0:
+    print(f"one-hot: {S.shape} nnz/rank={S.nnz} "
+          f"({S.nnz * 12 / 1e9:.2f} GB vs "
+          f"{S.shape[0] * S.shape[1] * 4 / 1e9:.1f} GB dense)")
+
+clf = LogisticRegression(C=1.0, max_iter=50).fit(S, from_numpy(y))
+acc = (clf.predict(S).gather().cpu().numpy() == y).mean()
+if comm.rank == 0:
+    print(f"train accuracy: {acc:.4f}")
+    print("feature names:", enc.get_feature_names_out()[:4], "...")
+
+comm_mod.shutdown()
