@@ -56,8 +56,25 @@ def kmeans_assign(x: torch.Tensor, centers: torch.Tensor
             x.contiguous(), centers.contiguous())
         return labels, sums, counts, inertia
     labels, mindist = closest_center(x, centers)
+    # deterministic centroid sums: label-sorted segment reduction per row
+    # block in fixed block order (index_add_ is atomic f64 on GPU and
+    # would break the bitwise run-to-run guarantee on this path)
     sums = torch.zeros(k, d, dtype=torch.float64, device=x.device)
-    sums.index_add_(0, labels.long(), x.to(torch.float64))
+    m = x.shape[0]
+    block = 4_194_304
+    for lo in range(0, m, block):
+        hi = min(m, lo + block)
+        lb = labels[lo:hi].long()
+        order = torch.argsort(lb, stable=True)
+        ls = lb[order]
+        uniq, cnt = torch.unique_consecutive(ls, return_counts=True)
+        offsets = torch.zeros(uniq.shape[0] + 1, dtype=torch.int64,
+                              device=x.device)
+        offsets[1:] = cnt.cumsum(0)
+        xs = x[lo:hi].to(torch.float64)[order]
+        part = torch.segment_reduce(xs, "sum", offsets=offsets, axis=0,
+                                    initial=0)
+        sums[uniq] += part
     counts = torch.bincount(labels.long(), minlength=k).to(torch.float64)
     inertia = mindist.to(torch.float64).sum()
     return labels, sums, counts, inertia
