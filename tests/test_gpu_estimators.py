@@ -167,3 +167,15 @@ def test_sparse_xtdot_bitwise_deterministic_gpu():
     z1 = S.dot(torch.ones(300, device="cuda")).gather().cpu().numpy()
     z2 = S.dot(torch.ones(300, device="cuda")).gather().cpu().numpy()
     assert (z1 == z2).all()
+
+
+def test_kmeans_large_k_fallback_deterministic_gpu():
+    from mi355ml.ops.kmeans import kmeans_assign
+    g = torch.Generator(device="cuda").manual_seed(7)
+    x = torch.randn(3_000_000, 32, generator=g, device="cuda")
+    c = torch.randn(400, 32, generator=g, device="cuda")  # k>256: composed
+    l1, s1, n1, i1 = kmeans_assign(x, c)
+    l2, s2, n2, i2 = kmeans_assign(x, c)
+    assert (l1 == l2).all()
+    assert (s1 == s2).all() and (n1 == n2).all()
+    assert float(i1) == float(i2)
