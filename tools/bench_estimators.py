@@ -66,7 +66,7 @@ def main():
                                           n_features=50, n_informative=10,
                                           random_state=3)
     with mi355ml.config.set({"array.block_rows": 100_000}):
-        hb = timeit("Hyperband(SGDClassifier, max_iter=81 brackets) 1e6x50",
+        hb = timeit("Hyperband(SGDClassifier CPU models, max_iter=27) 1e6x50",
                     lambda: HyperbandSearchCV(
                         SGDClassifier(loss="log_loss", random_state=0),
                         {"alpha": [10 ** -i for i in range(2, 7)],
