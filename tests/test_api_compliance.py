@@ -20,13 +20,27 @@ from mi355ml.preprocessing import (BlockTransformer, LabelEncoder,
                                    OrdinalEncoder, PolynomialFeatures,
                                    QuantileTransformer, RobustScaler,
                                    StandardScaler)
+from mi355ml.model_selection import (GridSearchCV, HyperbandSearchCV,
+                                     IncrementalSearchCV,
+                                     RandomizedSearchCV,
+                                     SuccessiveHalvingSearchCV)
+from mi355ml.wrappers import Incremental, ParallelPostFit
 
 ALL_ESTIMATORS = [
     KMeans(), SpectralClustering(), PCA(), TruncatedSVD(), IncrementalPCA(),
     SimpleImputer(), LinearRegression(), LogisticRegression(),
     PoissonRegression(), GaussianNB(), MinMaxScaler(), Normalizer(),
-    OneHotEncoder(), OrdinalEncoder(), PolynomialFeatures(),
-    QuantileTransformer(), RobustScaler(), StandardScaler(), LabelEncoder(),
+    OneHotEncoder(), OneHotEncoder(sparse_output=True, drop="first"),
+    OrdinalEncoder(), PolynomialFeatures(),
+    QuantileTransformer(), QuantileTransformer(method="tdigest"),
+    RobustScaler(), RobustScaler(method="tdigest"), StandardScaler(),
+    LabelEncoder(),
+    GridSearchCV(LogisticRegression(), {"C": [1.0]}),
+    RandomizedSearchCV(LogisticRegression(), {"C": [1.0, 2.0]}, n_iter=2),
+    HyperbandSearchCV(LogisticRegression(), {"C": [1.0, 2.0]}, max_iter=9),
+    SuccessiveHalvingSearchCV(LogisticRegression(), {"C": [1.0, 2.0]}),
+    IncrementalSearchCV(LogisticRegression(), {"C": [1.0, 2.0]}),
+    Incremental(LogisticRegression()), ParallelPostFit(LogisticRegression()),
 ]
 
 
