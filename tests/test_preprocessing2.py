@@ -229,3 +229,30 @@ def test_split_no_shuffle_and_kfold_shuffle():
         assert 150 < te_np.sum() < 350  # ~250 each in expectation
         seen += te_np
     np.testing.assert_array_equal(seen, 1.0)  # every row in exactly 1 fold
+
+
+def test_label_encoder_categorical_fast_path():
+    import pandas as pd
+    s = pd.Series(pd.Categorical(["b", "a", "c", "a", "b"],
+                                 categories=["a", "b", "c"]))
+    le = LabelEncoder().fit(s)
+    np.testing.assert_array_equal(le.classes_, ["a", "b", "c"])
+    codes = le.transform(s)
+    np.testing.assert_array_equal(codes, [1, 0, 2, 0, 1])
+    back = le.inverse_transform(codes)
+    assert list(back) == ["b", "a", "c", "a", "b"]
+    # unseen label -> error
+    s2 = pd.Series(pd.Categorical(["a", "z"], categories=["a", "z"]))
+    with pytest.raises(ValueError, match="unseen"):
+        le.transform(s2)
+    # DistList of categorical partitions
+    from mi355ml.core.bag import DistList
+    dl = DistList([s[:3], s[3:]])
+    le2 = LabelEncoder().fit(dl)
+    np.testing.assert_array_equal(le2.classes_, ["a", "b", "c"])
+    out = le2.transform(dl).gather()
+    np.testing.assert_array_equal(np.concatenate(out), [1, 0, 2, 0, 1])
+    # mismatched partition categories refuse (run Categorizer first)
+    bad = DistList([s[:3], pd.Series(pd.Categorical(["x"]))])
+    with pytest.raises(ValueError, match="mismatched"):
+        LabelEncoder().fit(bad)
