@@ -156,6 +156,8 @@ class SparseDistArray(DistArray):
 
     def sum(self, axis: Optional[int] = 0) -> torch.Tensor:
         """Per-column value sums (category frequencies), replicated [d]."""
+        if axis not in (0, None):
+            raise NotImplementedError("SparseDistArray.sum supports axis=0")
         ones = torch.ones(self.local_rows, dtype=torch.float64,
                           device=self.device)
         return self.xt_dot(DistArray(ones, self.global_rows,

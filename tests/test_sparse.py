@@ -3,6 +3,7 @@
 
 import numpy as np
 import scipy.sparse as sp
+import pytest
 import torch
 from sklearn.preprocessing import OneHotEncoder as SkOneHot
 
@@ -84,3 +85,15 @@ def _sparse_world2(rank, world, a_bytes, shape):
 def test_sparse_world2(mp_runner):
     a = _cat_data(800, 2, 4, seed=4)
     mp_runner(_sparse_world2, world=2, args=(a.tobytes(), a.shape))
+
+
+def test_sparse_multivector_dot():
+    rng = np.random.default_rng(5)
+    M = sp.random(300, 25, density=0.15, random_state=6, format="csr",
+                  dtype=np.float64)
+    S = from_scipy(M)
+    W = rng.standard_normal((25, 4))
+    Z = S.dot(torch.from_numpy(W)).gather().cpu().numpy()
+    np.testing.assert_allclose(Z, M @ W, rtol=1e-12)
+    with pytest.raises(NotImplementedError):
+        S.sum(axis=1)
