@@ -39,9 +39,12 @@ def _mp_entry(rank, world, port, fn, args):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     import mi355ml.parallel.comm as comm
-    comm.init(backend="gloo")
+    c = comm.init(backend="gloo")
     try:
         fn(rank, world, *args)
+        # a rank that finishes early must not exit (killing the rank-0
+        # TCPStore) while peers still use control-plane primitives
+        c.barrier()
     finally:
         comm.shutdown()
 

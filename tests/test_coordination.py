@@ -102,3 +102,31 @@ def _comm_counters_world2(rank, world):
 
 def test_comm_counters_world2(mp_runner):
     mp_runner(_comm_counters_world2, world=2, args=())
+
+
+def _joblib_ranks_world2(rank, world):
+    import os
+    import numpy as np
+    from joblib import Parallel, delayed, parallel_backend
+    import mi355ml.joblib_backend as jb
+    from mi355ml.parallel import comm as comm_mod
+    comm_mod.default()
+    jb.register()
+
+    def task(i):
+        return (i * 10, os.getpid())
+
+    with parallel_backend("mi355ml_ranks"):
+        out = Parallel()(delayed(task)(i) for i in range(8))
+    vals = [v for v, _ in out]
+    assert vals == [i * 10 for i in range(8)]
+    # tasks really ran in TWO different processes (round-robin ownership)
+    pids = {p for _, p in out}
+    assert len(pids) == 2
+    # every rank sees the same pids (results were broadcast from owners)
+    gathered = comm_mod.default().all_gather_object(sorted(pids))
+    assert gathered[0] == gathered[1]
+
+
+def test_joblib_rank_fanout_world2(mp_runner):
+    mp_runner(_joblib_ranks_world2, world=2, args=())
