@@ -180,3 +180,21 @@ def _shuffle_world2(rank, world, n):
 
 def test_dataframe_shuffle_world2(mp_runner):
     mp_runner(_shuffle_world2, world=2, args=(400,))
+
+
+def _merge_world2(rank, world, n):
+    from mi355ml.dataframe import concat, from_pandas, merge
+    rng = np.random.default_rng(6)
+    left = pd.DataFrame({"k": rng.integers(0, 9, n),
+                         "v": rng.standard_normal(n)})
+    right = pd.DataFrame({"k": np.arange(9), "w": np.arange(9) * 10})
+    got = concat(merge(from_pandas(left, npartitions=3),
+                       from_pandas(right, npartitions=2), on="k"))
+    exp = left.merge(right, on="k")
+    got = got.sort_values(["k", "v"]).reset_index(drop=True)
+    exp = exp.sort_values(["k", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_dataframe_merge_world2(mp_runner):
+    mp_runner(_merge_world2, world=2, args=(300,))

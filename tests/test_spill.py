@@ -87,3 +87,31 @@ def test_spilled_percentiles_and_robust_scaler():
     rs2 = RobustScaler().fit(X)
     np.testing.assert_allclose(rs.center_.cpu().numpy(),
                                rs2.center_.cpu().numpy(), atol=0.12)
+
+
+def _spilled_world2(rank, world, a_bytes, y_bytes, n, f):
+    X = np.frombuffer(a_bytes, dtype=np.float32).reshape(n, f).copy()
+    y = np.frombuffer(y_bytes, dtype=np.float32).copy()
+    S = spill_rows(torch.from_numpy(X), block_rows=97)
+    # streamed moments equal the resident distributed moments
+    D = from_numpy(X)
+    ns, ms, m2s = S.moments()
+    nr, mr, m2r = D.moments()
+    assert ns == nr
+    np.testing.assert_allclose(ms.cpu().numpy(), mr.cpu().numpy(), rtol=1e-10)
+    np.testing.assert_allclose(m2s.cpu().numpy(), m2r.cpu().numpy(),
+                               rtol=1e-8)
+    # spilled fit == resident fit across 2 ranks
+    lr_s = LogisticRegression(max_iter=40).fit(S, from_numpy(y))
+    lr_d = LogisticRegression(max_iter=40).fit(D, from_numpy(y))
+    np.testing.assert_allclose(lr_s.coef_.cpu().numpy(),
+                               lr_d.coef_.cpu().numpy(), rtol=1e-5,
+                               atol=1e-7)
+
+
+def test_spilled_world2(mp_runner):
+    rng = np.random.default_rng(21)
+    X = rng.standard_normal((1500, 6)).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] > 0).astype(np.float32)
+    mp_runner(_spilled_world2, world=2,
+              args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1]))
