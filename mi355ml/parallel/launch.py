@@ -34,12 +34,18 @@ def main(argv=None):
             n = 1
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     from torch.distributed.run import main as torchrun
+    extra = []
+    if args.max_restarts > 0:
+        # fast failure detection: the default 5 s monitor interval delays
+        # the respawn well past the surviving ranks' comm timeout
+        extra.append("--monitor-interval=1")
     torchrun([
         "--nnodes=1",
         f"--nproc-per-node={n}",
         "--master-addr=127.0.0.1",
         f"--master-port={args.master_port}",
         f"--max-restarts={args.max_restarts}",
+        *extra,
         args.script,
         *args.script_args,
     ])
