@@ -17,7 +17,7 @@ def _split(arr, cuts):
     return [arr[a:b] for a, b in zip(idx[:-1], idx[1:])]
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 @given(
     n=st.integers(20, 400),
     f=st.integers(1, 5),
@@ -45,7 +45,7 @@ def test_moment_merge_partition_invariant(n, f, seed, cuts, scale):
                                rtol=1e-8, atol=1e-10 * scale * scale)
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=30, deadline=None, derandomize=True)
 @given(
     n=st.integers(50, 500),
     seed=st.integers(0, 10_000),
@@ -68,17 +68,27 @@ def test_percentile_merge_bounds_and_monotone(n, seed, cuts, qs):
     merged = merge_percentiles(qs, torch.stack(curves),
                                torch.tensor(ns, dtype=torch.float64), q_ext)
     vals = merged[:, 0].numpy()
-    # monotone in q and inside the data range
+    # invariants of the weighted-CDF merge:
+    # 1) monotone in q and inside the data range
     assert (np.diff(vals) >= -1e-12).all()
     assert vals.min() >= x.min() - 1e-9 and vals.max() <= x.max() + 1e-9
-    # each estimate lands within a generous rank window of the target
-    order = np.sort(x)
+    # 2) each merged q-quantile is bracketed by the per-part q-quantiles
+    #    (a part with everything below/above pulls the merge toward it,
+    #    but never past the extreme part)
+    per_part = np.stack([c.numpy()[:, 0] for c in curves])  # [parts, nq_ext]
+    qe = q_ext.numpy()
     for q, v in zip(qs, vals):
-        r = np.searchsorted(order, v) / n * 100
-        assert abs(r - q) < 100.0 / max(len(curves), 1) + 35
+        j = int(np.where(qe == q)[0][0])
+        lo = per_part[:, j].min()
+        hi = per_part[:, j].max()
+        assert lo - 1e-9 <= v <= hi + 1e-9
+    # 3) single partition -> exact quantiles
+    if len(curves) == 1:
+        exact = np.quantile(x, np.asarray(qs) / 100.0)
+        np.testing.assert_allclose(vals, exact, rtol=1e-9, atol=1e-9)
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=25, deadline=None, derandomize=True)
 @given(
     n=st.integers(200, 3000),
     seed=st.integers(0, 10_000),
@@ -110,7 +120,7 @@ def test_tdigest_merge_rank_accuracy(n, seed, nparts):
         assert abs(r - q) < 3.0  # t-digest rank error bound (delta=500)
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=20, deadline=None, derandomize=True)
 @given(seed=st.integers(0, 2**31 - 1))
 def test_shard_seed_streams_distinct(seed):
     from mi355ml.core.random import shard_seed
