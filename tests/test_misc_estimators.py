@@ -152,3 +152,13 @@ def test_text_pipeline_sparse_incremental():
     preds = inc.predict(Xs).gather()
     pred = np.concatenate([np.asarray(p) for p in preds])
     assert (pred == labels).mean() > 0.9
+
+
+def test_xgboost_handoff_informative_error():
+    import pytest as _pytest
+    from mi355ml import xgboost as mx
+    from mi355ml.core.array import from_numpy
+    X = from_numpy(np.zeros((10, 2), dtype=np.float32))
+    y = from_numpy(np.zeros(10, dtype=np.float32))
+    with _pytest.raises(ImportError, match="xgboost"):
+        mx.train({}, X, y)
