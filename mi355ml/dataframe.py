@@ -96,9 +96,14 @@ def shuffle(frames: DistList, on: Union[str, List[str]]) -> DistList:
     (the reference's rearrange_by_column task shuffle,
     dask/dataframe/shuffle.py:443). Enables exact groupby-apply and joins.
 
-    Dataframe partitions are control-plane data here (the dense data plane
-    is DistArray), so the exchange is one object all-gather; each rank
-    keeps the slices hashed to it."""
+    Exchange is true point-to-point (the task-shuffle data plane): each
+    (src, dst) pair moves exactly its pickled slice with one isend/irecv
+    over the direct link — not a W-fold object all-gather."""
+    import pickle
+
+    import torch
+    import torch.distributed as dist
+
     comm = frames.comm
     W = comm.world_size
     if W == 1:
@@ -117,13 +122,38 @@ def shuffle(frames: DistList, on: Union[str, List[str]]) -> DistList:
                 outgoing[r].append(sel)
     template = next((p.iloc[:0] for p in frames.local
                      if p is not None and p.shape[1]), None)
-    gathered = comm.all_gather_object(
-        ([pd.concat(o) if o else None for o in outgoing], template))
-    mine = [out[comm.rank] for out, _ in gathered
-            if out[comm.rank] is not None]
+    blobs = [pickle.dumps(pd.concat(o)) if o else b"" for o in outgoing]
+    # size matrix via one small all-gather, then batched P2P of the blobs
+    sizes_local = torch.tensor([len(b) for b in blobs], dtype=torch.int64)
+    sizes = [s.cpu() for s in comm.all_gather(sizes_local.to(comm.device)
+                                              if comm.device.type == "cuda"
+                                              else sizes_local)]
+    ops = []
+    recv_bufs = {}
+    for r in range(W):
+        if r == comm.rank:
+            continue
+        if len(blobs[r]):
+            ops.append(dist.P2POp(dist.isend, torch.frombuffer(
+                bytearray(blobs[r]), dtype=torch.uint8), r))
+        n_in = int(sizes[r][comm.rank])
+        if n_in:
+            buf = torch.empty(n_in, dtype=torch.uint8)
+            recv_bufs[r] = buf
+            ops.append(dist.P2POp(dist.irecv, buf, r))
+    if ops:
+        for req in dist.batch_isend_irecv(ops):
+            req.wait()
+    mine = []
+    if blobs[comm.rank]:
+        mine.append(pickle.loads(blobs[comm.rank]))
+    for r in sorted(recv_bufs):
+        mine.append(pickle.loads(recv_bufs[r].numpy().tobytes()))
+    # template exchange keeps empty ranks schema-correct for merges
+    templates = comm.all_gather_object(template)
     if mine:
         return DistList([pd.concat(mine)], comm)
-    empty = next((t for _, t in gathered if t is not None), pd.DataFrame())
+    empty = next((t for t in templates if t is not None), pd.DataFrame())
     return DistList([empty], comm)
 
 
