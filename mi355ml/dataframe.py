@@ -166,7 +166,12 @@ def groupby_apply(frames: DistList, by: Union[str, List[str]],
     def _apply(part: pd.DataFrame):
         if part is None or not len(part):
             return part
-        return part.groupby(by, group_keys=False).apply(fn)
+        # explicit per-group loop: fn sees the FULL sub-frame including the
+        # key columns (dask groupby.apply semantics), independent of the
+        # pandas include_groups deprecation cycle
+        outs = [fn(sub) for _, sub in part.groupby(by, group_keys=False,
+                                                   sort=True)]
+        return pd.concat(outs) if outs else part.iloc[:0]
 
     return shuffled.map(_apply)
 

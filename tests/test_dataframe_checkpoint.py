@@ -138,8 +138,8 @@ def test_dataframe_shuffle_groupby_apply():
     # groupby-apply: demean within each key group (needs co-located groups)
     out = groupby_apply(frames, "k", lambda g: g.assign(v=g.v - g.v.mean()))
     got = concat(out)
-    exp2 = df.groupby("k", group_keys=False).apply(
-        lambda g: g.assign(v=g.v - g.v.mean()))
+    exp2 = pd.concat([g.assign(v=g.v - g.v.mean())
+                      for _, g in df.groupby("k", group_keys=False)])
     got = got.sort_values(["k", "v"]).reset_index(drop=True)
     exp2 = exp2.sort_values(["k", "v"]).reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp2)
