@@ -101,3 +101,24 @@ def test_pca_world2(mp_runner):
     mp_runner(_pca_world2, world=2,
               args=(X.tobytes(), X.shape[0], X.shape[1],
                     ref.singular_values_))
+
+
+def test_pca_whiten_matches_sklearn():
+    from sklearn.decomposition import PCA as SkPCA
+    rng = np.random.default_rng(17)
+    B = rng.standard_normal((5, 12))
+    X = (rng.standard_normal((4000, 5)) @ B
+         + 0.05 * rng.standard_normal((4000, 12))).astype(np.float32)
+    ours = PCA(n_components=4, whiten=True, svd_solver="full").fit(
+        from_numpy(X))
+    ref = SkPCA(n_components=4, whiten=True).fit(X)
+    Z = ours.transform(from_numpy(X)).gather().cpu().numpy()
+    Zr = ref.transform(X)
+    # whitened components have unit variance; sign per component is free
+    np.testing.assert_allclose(Z.std(axis=0), Zr.std(axis=0), rtol=1e-2)
+    cos = np.abs((Z * Zr).mean(axis=0) / (Z.std(axis=0) * Zr.std(axis=0)))
+    assert (cos > 0.99).all()
+    # inverse_transform roundtrip
+    back = ours.inverse_transform(from_numpy(Z)).gather().cpu().numpy()
+    np.testing.assert_allclose(back, ref.inverse_transform(Zr), rtol=1e-2,
+                               atol=1e-2)

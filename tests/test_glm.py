@@ -99,3 +99,22 @@ def test_logreg_world2(mp_runner):
     mp_runner(_logreg_world2, world=2,
               args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1],
                     ref.coef_[0]))
+
+
+def test_poisson_partial_fit_improves():
+    rng = np.random.default_rng(23)
+    X = rng.standard_normal((4000, 5)).astype(np.float32)
+    w = np.array([0.4, -0.3, 0.2, 0.0, 0.1])
+    y = rng.poisson(np.exp(X @ w)).astype(np.float32)
+    pr = PoissonRegression(eta0=0.05)
+    from mi355ml.core.array import from_numpy as fn
+    losses = []
+    from mi355ml.glm.algorithms import _dist_value_grad
+    import torch
+    for ep in range(30):
+        pr.partial_fit(fn(X), fn(y))
+        l, _, _ = _dist_value_grad(fn(X), fn(y),
+                                   pr.coef_, pr.intercept_, "poisson")
+        losses.append(l)
+    assert losses[-1] < losses[0] - 1.0  # SGD steps reduce the deviance
+    np.testing.assert_allclose(pr.coef_.cpu().numpy(), w, atol=0.25)

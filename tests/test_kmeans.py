@@ -82,3 +82,17 @@ def test_kmeans_world2(mp_runner):
     ref = SkKMeans(n_clusters=4, n_init=10, random_state=0).fit(X)
     mp_runner(_kmeans_world2, world=2,
               args=(X.tobytes(), X.shape[0], X.shape[1], ref.inertia_))
+
+
+def test_kmeans_explicit_init_matrix():
+    rng = np.random.default_rng(19)
+    centers = rng.uniform(-6, 6, (3, 4))
+    lab = rng.integers(0, 3, 900)
+    X = (centers[lab] + 0.3 * rng.standard_normal((900, 4))).astype(
+        np.float32)
+    km = KMeans(n_clusters=3, init=centers.astype(np.float32),
+                max_iter=10).fit(from_numpy(X))
+    got = km.cluster_centers_.cpu().numpy()
+    # each true center recovered to within the noise scale
+    d = np.linalg.norm(got[:, None, :] - centers[None, :, :], axis=-1)
+    assert d.min(axis=0).max() < 0.2
