@@ -162,3 +162,27 @@ def test_xgboost_handoff_informative_error():
     y = from_numpy(np.zeros(10, dtype=np.float32))
     with _pytest.raises(ImportError, match="xgboost"):
         mx.train({}, X, y)
+
+
+def test_metrics_match_sklearn():
+    import sklearn.metrics as skm
+    from mi355ml.metrics import (accuracy_score, log_loss,
+                                 mean_absolute_error, mean_squared_error,
+                                 r2_score)
+    rng = np.random.default_rng(29)
+    yt = rng.integers(0, 2, 2000).astype(np.float32)
+    yp = rng.integers(0, 2, 2000).astype(np.float32)
+    proba = rng.uniform(0.01, 0.99, 2000).astype(np.float32)
+    p2 = np.stack([1 - proba, proba], axis=1)
+    from mi355ml.core.array import from_numpy
+    assert abs(float(accuracy_score(from_numpy(yt), from_numpy(yp)))
+               - skm.accuracy_score(yt, yp)) < 1e-6
+    assert abs(float(log_loss(from_numpy(yt), from_numpy(p2)))
+               - skm.log_loss(yt, p2)) < 1e-5
+    yr = rng.standard_normal(2000).astype(np.float32)
+    pr = (yr + 0.3 * rng.standard_normal(2000)).astype(np.float32)
+    for ours, ref in ((mean_squared_error, skm.mean_squared_error),
+                      (mean_absolute_error, skm.mean_absolute_error),
+                      (r2_score, skm.r2_score)):
+        assert abs(float(ours(from_numpy(yr), from_numpy(pr)))
+                   - ref(yr, pr)) < 1e-4
