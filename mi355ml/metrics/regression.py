@@ -54,6 +54,8 @@ def r2_score(y_true, y_pred) -> float:
     ss_res, s, s2, n = _reduce_sums(yt.comm, ss_res, s, s2, n)
     mean = s / max(n, 1.0)
     ss_tot = s2 - n * mean * mean
-    if ss_tot <= 0:
-        return 0.0
+    if ss_tot <= 1e-12 * max(s2, 1.0):
+        # constant y_true: sklearn returns 1.0 for a perfect prediction,
+        # 0.0 otherwise (advisor r1 low)
+        return 1.0 if ss_res <= 1e-12 * max(s2, float(n)) else 0.0
     return 1.0 - ss_res / ss_tot

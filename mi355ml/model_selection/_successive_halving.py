@@ -34,11 +34,15 @@ class SuccessiveHalvingSearchCV(BaseIncrementalSearchCV):
         if len(alive) == 1:
             keep = 1
         survivors = alive[:keep]
-        # stop when a single survivor has exhausted max_iter (or no growth)
+        # Rung budget r_i = r0*eta^rung is the TARGET TOTAL number of
+        # partial_fit calls after the rung (reference semantics), so each
+        # survivor is granted r_i - pf_calls additional calls, clamped to
+        # max_iter (advisor r1 medium: granting r_i as *extra* calls
+        # inflated training cost ~eta/(eta-1)x).
         budget = self._budget(round_idx)
         out = {}
         for t in survivors:
-            extra = min(budget, self.max_iter - t.pf_calls)
+            extra = min(budget, self.max_iter) - t.pf_calls
             if extra > 0:
                 out[t.tid] = extra
         if len(survivors) <= 1 and not out:

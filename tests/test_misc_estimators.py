@@ -186,3 +186,11 @@ def test_metrics_match_sklearn():
                       (r2_score, skm.r2_score)):
         assert abs(float(ours(from_numpy(yr), from_numpy(pr)))
                    - ref(yr, pr)) < 1e-4
+
+
+def test_r2_constant_y_true_edge_cases():
+    # sklearn: perfect prediction of a constant y -> 1.0, otherwise 0.0
+    from mi355ml.metrics import r2_score
+    y = np.full(64, 3.0, dtype=np.float32)
+    assert r2_score(from_numpy(y), from_numpy(y.copy())) == 1.0
+    assert r2_score(from_numpy(y), from_numpy(y + 1.0)) == 0.0
