@@ -56,7 +56,8 @@ class _SearchBase(BaseEstimator):
     _SMALL_DATA_CELLS = 2_000_000
 
     def __init__(self, estimator, scoring=None, cv=None, refit=True,
-                 return_train_score=False, candidate_parallel="auto"):
+                 return_train_score=False, candidate_parallel="auto",
+                 error_score=np.nan):
         self.estimator = estimator
         self.scoring = scoring
         self.cv = cv
@@ -67,47 +68,87 @@ class _SearchBase(BaseEstimator):
         # reference's cluster-wide candidate fan-out); "ranks" forces it;
         # "none" keeps every candidate a distributed fit.
         self.candidate_parallel = candidate_parallel
+        # sklearn semantics: "raise" re-raises a failing candidate fit;
+        # a number is recorded as that candidate's score (with a warning)
+        self.error_score = error_score
 
     def _candidates(self) -> List[Dict]:
         raise NotImplementedError
 
-    def _score(self, est, Xte, yte) -> float:
-        if self.scoring is not None:
-            from ..metrics.scorer import check_scoring
-            scorer = check_scoring(est, self.scoring)
-            return scorer(est, Xte, yte)
-        return est.score(Xte, yte) if yte is not None else est.score(Xte)
+    def _scorers(self):
+        """Normalize self.scoring into ({name: scorer}, multimetric).
+        Single metric keeps the legacy name 'score' (sklearn parity)."""
+        from ..metrics.scorer import check_scoring, get_scorer
+        s = self.scoring
+        if s is None or isinstance(s, str) or callable(s):
+            return {"score": check_scoring(self.estimator, s)}, False
+        if isinstance(s, (list, tuple, set)):
+            return {name: get_scorer(name) for name in s}, True
+        if isinstance(s, dict):
+            return {name: get_scorer(sc) for name, sc in s.items()}, True
+        raise ValueError(f"unsupported scoring spec {s!r}")
 
-    def _fit_candidate(self, params, Xtr, ytr, Xte, yte, cache, fold):
-        est = clone(self.estimator)
-        est.set_params(**params)
-        if _is_pipeline(est) and len(est.steps) > 1:
-            # pipeline-prefix dedup: fit_transform shared prefixes once
-            Xtr_c, Xte_c = Xtr, Xte
-            prefix_token = str(fold)
-            steps = est.steps
-            for name, step in steps[:-1]:
-                prefix_token = _tokenize([prefix_token, name, step])
-                hit = cache.get(prefix_token)
-                if hit is None:
-                    stf = clone(step).fit(Xtr_c, ytr)
-                    Xtr_n = stf.transform(Xtr_c)
-                    Xte_n = stf.transform(Xte_c)
-                    cache[prefix_token] = (stf, Xtr_n, Xte_n)
-                stf, Xtr_c, Xte_c = cache[prefix_token]
-            final_name, final = steps[-1]
-            finalf = clone(final)
-            if ytr is not None:
-                finalf.fit(Xtr_c, ytr)
+    def _score(self, est, Xte, yte) -> float:
+        scorers, multi = self._scorers()
+        if multi:
+            name = self.refit if isinstance(self.refit, str) else \
+                next(iter(scorers))
+            return scorers[name](est, Xte, yte)
+        return scorers["score"](est, Xte, yte)
+
+    def _fit_candidate(self, params, Xtr, ytr, Xte, yte, cache, fold,
+                       scorers):
+        """Fit one candidate on one fold; returns ({name: test_score},
+        {name: train_score} or None). A failing fit/score records
+        self.error_score (sklearn semantics) unless it is 'raise'."""
+        want_train = self.return_train_score
+        try:
+            est = clone(self.estimator)
+            est.set_params(**params)
+            if _is_pipeline(est) and len(est.steps) > 1:
+                # pipeline-prefix dedup: fit_transform shared prefixes once
+                Xtr_c, Xte_c = Xtr, Xte
+                prefix_token = str(fold)
+                steps = est.steps
+                for name, step in steps[:-1]:
+                    prefix_token = _tokenize([prefix_token, name, step])
+                    hit = cache.get(prefix_token)
+                    if hit is None:
+                        stf = clone(step).fit(Xtr_c, ytr)
+                        Xtr_n = stf.transform(Xtr_c)
+                        Xte_n = stf.transform(Xte_c)
+                        cache[prefix_token] = (stf, Xtr_n, Xte_n)
+                    stf, Xtr_c, Xte_c = cache[prefix_token]
+                final_name, final = steps[-1]
+                fitted = clone(final)
+                if ytr is not None:
+                    fitted.fit(Xtr_c, ytr)
+                else:
+                    fitted.fit(Xtr_c)
+                Xte_s, Xtr_s = Xte_c, Xtr_c
             else:
-                finalf.fit(Xtr_c)
-            score = self._score(finalf, Xte_c, yte)
-            return score
-        if ytr is not None:
-            est.fit(Xtr, ytr)
-        else:
-            est.fit(Xtr)
-        return self._score(est, Xte, yte)
+                fitted = est
+                if ytr is not None:
+                    fitted.fit(Xtr, ytr)
+                else:
+                    fitted.fit(Xtr)
+                Xte_s, Xtr_s = Xte, Xtr
+            test = {n: float(sc(fitted, Xte_s, yte))
+                    for n, sc in scorers.items()}
+            train = ({n: float(sc(fitted, Xtr_s, ytr))
+                      for n, sc in scorers.items()} if want_train else None)
+            return test, train
+        except Exception:
+            if self.error_score == "raise":
+                raise
+            import warnings
+            warnings.warn(
+                f"candidate {params!r} failed on fold {fold}; scoring it "
+                f"error_score={self.error_score!r}", UserWarning)
+            err = float(self.error_score)
+            test = {n: err for n in scorers}
+            train = {n: err for n in scorers} if want_train else None
+            return test, train
 
     def _use_candidate_parallel(self, X, n_candidates: int) -> bool:
         if X.comm.world_size == 1 or n_candidates < 2:
@@ -119,9 +160,12 @@ class _SearchBase(BaseEstimator):
                     <= self._SMALL_DATA_CELLS)
         return False
 
-    def _score_grid(self, X, y, candidates, cv, n_splits) -> np.ndarray:
-        """scores[c, f] with every candidate fitted by ALL ranks."""
-        scores = np.zeros((len(candidates), n_splits))
+    def _score_grid(self, X, y, candidates, cv, n_splits, scorers):
+        """scores[name][c, f] with every candidate fitted by ALL ranks."""
+        C = len(candidates)
+        test = {n: np.full((C, n_splits), np.nan) for n in scorers}
+        train = ({n: np.full((C, n_splits), np.nan) for n in scorers}
+                 if self.return_train_score else None)
         for f, (tr_mask, te_mask) in enumerate(cv.split(X, y)):
             arrays = [X] + ([y] if y is not None else [])
             tr = mask_take(arrays, tr_mask)
@@ -131,12 +175,16 @@ class _SearchBase(BaseEstimator):
             yte = te[1] if y is not None else None
             cache: Dict[str, Any] = {}
             for c, params in enumerate(candidates):
-                scores[c, f] = self._fit_candidate(
-                    params, Xtr, ytr, Xte, yte, cache, f)
-        return scores
+                ts, tr_s = self._fit_candidate(
+                    params, Xtr, ytr, Xte, yte, cache, f, scorers)
+                for n in scorers:
+                    test[n][c, f] = ts[n]
+                    if train is not None:
+                        train[n][c, f] = tr_s[n]
+        return test, train
 
-    def _score_grid_rank_parallel(self, X, y, candidates, cv, n_splits
-                                  ) -> np.ndarray:
+    def _score_grid_rank_parallel(self, X, y, candidates, cv, n_splits,
+                                  scorers):
         """Candidates fan out round-robin across ranks; each rank fits its
         subset on the locally-gathered (small) data with a single-process
         communicator, then scores are all-gathered (the reference's
@@ -148,7 +196,10 @@ class _SearchBase(BaseEstimator):
         yl = (DistArray(y.gather(), y.global_rows, 0, lcomm)
               if y is not None else None)
         mine = list(range(comm.rank, len(candidates), comm.world_size))
-        scores = np.full((len(candidates), n_splits), np.nan)
+        C = len(candidates)
+        test = {n: np.full((C, n_splits), np.nan) for n in scorers}
+        train = ({n: np.full((C, n_splits), np.nan) for n in scorers}
+                 if self.return_train_score else None)
         for f, (tr_mask, te_mask) in enumerate(cv.split(Xl, yl)):
             arrays = [Xl] + ([yl] if yl is not None else [])
             tr = mask_take(arrays, tr_mask)
@@ -158,13 +209,28 @@ class _SearchBase(BaseEstimator):
             yte = te[1] if yl is not None else None
             cache: Dict[str, Any] = {}
             for c in mine:
-                scores[c, f] = self._fit_candidate(
-                    candidates[c], Xtr, ytr, Xte, yte, cache, f)
-        parts = comm.all_gather_object({c: scores[c].tolist() for c in mine})
-        for part in parts:
-            for c, row in part.items():
-                scores[c] = row
-        return scores
+                ts, tr_s = self._fit_candidate(
+                    candidates[c], Xtr, ytr, Xte, yte, cache, f, scorers)
+                for n in scorers:
+                    test[n][c, f] = ts[n]
+                    if train is not None:
+                        train[n][c, f] = tr_s[n]
+        payload = {c: {n: (test[n][c].tolist(),
+                           train[n][c].tolist() if train is not None
+                           else None) for n in scorers} for c in mine}
+        for part in comm.all_gather_object(payload):
+            for c, per_name in part.items():
+                for n, (trow, trrow) in per_name.items():
+                    test[n][c] = trow
+                    if train is not None and trrow is not None:
+                        train[n][c] = trrow
+        return test, train
+
+    @staticmethod
+    def _rank(mean: np.ndarray) -> np.ndarray:
+        """sklearn-style ranks, NaN (failed) candidates ranked last."""
+        key = np.where(np.isnan(mean), -np.inf, mean)
+        return ((-key).argsort(kind="stable").argsort() + 1).astype(int)
 
     def fit(self, X, y=None, **fit_params):
         X = check_array(X, dtype=None)
@@ -172,27 +238,46 @@ class _SearchBase(BaseEstimator):
             y = check_array(y, dtype=None)
         cv = check_cv(self.cv)
         candidates = self._candidates()
+        scorers, multimetric = self._scorers()
+        if multimetric and self.refit is not False and \
+                not (isinstance(self.refit, str) and self.refit in scorers):
+            raise ValueError(
+                "with multimetric scoring, refit must be False or the name "
+                f"of a metric in {sorted(scorers)}")
         n_splits = cv.get_n_splits(X, y)
         if self._use_candidate_parallel(X, len(candidates)):
-            scores = self._score_grid_rank_parallel(X, y, candidates, cv,
-                                                    n_splits)
+            test, train = self._score_grid_rank_parallel(
+                X, y, candidates, cv, n_splits, scorers)
         else:
-            scores = self._score_grid(X, y, candidates, cv, n_splits)
-        mean = scores.mean(axis=1)
-        std = scores.std(axis=1)
-        best = int(np.argmax(mean))
-        self.cv_results_ = {
-            "params": candidates,
-            "mean_test_score": mean,
-            "std_test_score": std,
-            "rank_test_score": ((-mean).argsort().argsort() + 1).astype(int),
-        }
-        for f in range(n_splits):
-            self.cv_results_[f"split{f}_test_score"] = scores[:, f]
+            test, train = self._score_grid(X, y, candidates, cv, n_splits,
+                                           scorers)
+        self.cv_results_ = {"params": candidates}
+        self.multimetric_ = multimetric
+        suffix = (lambda n: n) if multimetric else (lambda n: "score")
+        for n in scorers:
+            sn = suffix(n)
+            mean = test[n].mean(axis=1)
+            self.cv_results_[f"mean_test_{sn}"] = mean
+            self.cv_results_[f"std_test_{sn}"] = test[n].std(axis=1)
+            self.cv_results_[f"rank_test_{sn}"] = self._rank(mean)
+            for f in range(n_splits):
+                self.cv_results_[f"split{f}_test_{sn}"] = test[n][:, f]
+            if train is not None:
+                self.cv_results_[f"mean_train_{sn}"] = train[n].mean(axis=1)
+                self.cv_results_[f"std_train_{sn}"] = train[n].std(axis=1)
+                for f in range(n_splits):
+                    self.cv_results_[f"split{f}_train_{sn}"] = train[n][:, f]
+        refit_name = (self.refit if isinstance(self.refit, str)
+                      else next(iter(scorers)))
+        mean = self.cv_results_[f"mean_test_{suffix(refit_name)}"]
+        if np.isnan(mean).all():
+            raise ValueError("every candidate failed; cannot pick a best "
+                             "(error_score recorded NaN for all)")
+        best = int(np.nanargmax(mean))
         self.best_index_ = best
         self.best_params_ = candidates[best]
         self.best_score_ = float(mean[best])
-        if self.refit:
+        if self.refit is not False and self.refit is not None:
             est = clone(self.estimator)
             est.set_params(**self.best_params_)
             if y is not None:
@@ -220,9 +305,9 @@ class _SearchBase(BaseEstimator):
 class GridSearchCV(_SearchBase):
     def __init__(self, estimator, param_grid, scoring=None, cv=None,
                  refit=True, return_train_score=False,
-                 candidate_parallel="auto"):
+                 candidate_parallel="auto", error_score=np.nan):
         super().__init__(estimator, scoring, cv, refit, return_train_score,
-                         candidate_parallel)
+                         candidate_parallel, error_score)
         self.param_grid = param_grid
 
     def _candidates(self):
@@ -232,9 +317,10 @@ class GridSearchCV(_SearchBase):
 class RandomizedSearchCV(_SearchBase):
     def __init__(self, estimator, param_distributions, n_iter=10,
                  scoring=None, cv=None, refit=True, random_state=0,
-                 return_train_score=False, candidate_parallel="auto"):
+                 return_train_score=False, candidate_parallel="auto",
+                 error_score=np.nan):
         super().__init__(estimator, scoring, cv, refit, return_train_score,
-                         candidate_parallel)
+                         candidate_parallel, error_score)
         self.param_distributions = param_distributions
         self.n_iter = n_iter
         self.random_state = random_state

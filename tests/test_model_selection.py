@@ -283,3 +283,79 @@ def _skewed_scoring_world2(rank, world, n):
 
 def test_incremental_global_scoring_skewed_shards_world2(mp_runner):
     mp_runner(_skewed_scoring_world2, world=2, args=(1000,))
+
+
+def test_grid_search_return_train_score():
+    X, y = _cls(n=1500, seed=13)
+    gs = GridSearchCV(LogisticRegression(solver="lbfgs", max_iter=50),
+                      {"C": [0.1, 1.0]}, cv=3, return_train_score=True)
+    gs.fit(from_numpy(X), from_numpy(y))
+    assert "mean_train_score" in gs.cv_results_
+    assert "split0_train_score" in gs.cv_results_
+    tr = gs.cv_results_["mean_train_score"]
+    te = gs.cv_results_["mean_test_score"]
+    assert tr.shape == te.shape == (2,)
+    assert not np.isnan(tr).any()
+    # train score should be >= test score on average (same-data fit)
+    assert tr.mean() >= te.mean() - 0.05
+
+
+class _FailingEstimator:
+    """Fails to fit whenever C is negative (error_score machinery)."""
+
+    def __init__(self, C=1.0):
+        self.C = C
+
+    def get_params(self, deep=True):
+        return {"C": self.C}
+
+    def set_params(self, **p):
+        for k, v in p.items():
+            setattr(self, k, v)
+        return self
+
+    def fit(self, X, y=None):
+        if self.C < 0:
+            raise ValueError("boom")
+        self.mean_ = 0.0
+        return self
+
+    def score(self, X, y=None):
+        return 0.5 + 0.1 * self.C
+
+
+def test_grid_search_error_score():
+    X, y = _cls(n=400, f=4, seed=14)
+    gs = GridSearchCV(_FailingEstimator(), {"C": [-1.0, 1.0, 2.0]}, cv=2,
+                      error_score=-999.0, candidate_parallel="none")
+    with pytest.warns(UserWarning, match="failed on fold"):
+        gs.fit(from_numpy(X), from_numpy(y))
+    assert gs.cv_results_["mean_test_score"][0] == -999.0
+    assert gs.best_params_ == {"C": 2.0}
+    # failed candidate ranked last
+    assert gs.cv_results_["rank_test_score"][0] == 3
+    # error_score='raise' re-raises
+    gs2 = GridSearchCV(_FailingEstimator(), {"C": [-1.0, 1.0]}, cv=2,
+                       error_score="raise", candidate_parallel="none")
+    with pytest.raises(ValueError, match="boom"):
+        gs2.fit(from_numpy(X), from_numpy(y))
+
+
+def test_grid_search_multimetric():
+    X, y = _cls(n=1500, seed=15)
+    gs = GridSearchCV(LogisticRegression(solver="lbfgs", max_iter=50),
+                      {"C": [0.1, 1.0]},
+                      scoring={"acc": "accuracy", "nll": "neg_log_loss"},
+                      cv=3, refit="acc", return_train_score=True)
+    gs.fit(from_numpy(X), from_numpy(y))
+    for key in ("mean_test_acc", "mean_test_nll", "rank_test_acc",
+                "split0_test_nll", "mean_train_acc"):
+        assert key in gs.cv_results_, key
+    assert gs.multimetric_
+    assert gs.best_score_ == gs.cv_results_["mean_test_acc"][gs.best_index_]
+    assert hasattr(gs, "best_estimator_")
+    # refit must name a metric under multimetric scoring
+    bad = GridSearchCV(LogisticRegression(), {"C": [1.0]},
+                       scoring=["accuracy", "neg_log_loss"], cv=2)
+    with pytest.raises(ValueError, match="refit"):
+        bad.fit(from_numpy(X), from_numpy(y))
