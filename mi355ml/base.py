@@ -35,6 +35,20 @@ def check_array(X: ArrayLike, dtype: torch.dtype = torch.float32,
                                         X.row_offset, X.comm, X.block_rows)
             return X.astype(dtype)
         return X
+    from .core.colarray import ColDistArray, Replicated
+    if isinstance(X, ColDistArray):
+        # column-sharded (wide-X) arrays pass through untouched
+        if dtype is not None and X.dtype != dtype:
+            return X.astype(dtype)
+        return X
+    if isinstance(X, Replicated):
+        # replicated per-row result of a feature-parallel op: every rank
+        # holds the identical full tensor — slice it into a valid
+        # row-sharded DistArray
+        t = X.local
+        if dtype is not None and t.dtype != dtype:
+            t = t.to(dtype)
+        return shard_rows(t, comm or X.comm)
     if isinstance(X, np.ndarray):
         t = torch.from_numpy(np.ascontiguousarray(X))
     elif isinstance(X, torch.Tensor):
@@ -49,6 +63,24 @@ def check_array(X: ArrayLike, dtype: torch.dtype = torch.float32,
 def check_X_y(X: ArrayLike, y: ArrayLike, dtype: torch.dtype = torch.float32,
               y_dtype: torch.dtype = torch.float32, comm=None):
     Xd = check_array(X, dtype, comm)
+    from .core.colarray import ColDistArray, Replicated
+    if isinstance(Xd, ColDistArray):
+        # col-sharded X holds ALL rows on every rank: y must be the full
+        # replicated label vector, not a row shard
+        if isinstance(y, DistArray):
+            yt = y.gather()
+        elif isinstance(y, Replicated):
+            yt = y.local
+        elif isinstance(y, torch.Tensor):
+            yt = y
+        else:
+            yt = torch.as_tensor(np.asarray(y))
+        yt = yt.to(device=Xd.device, dtype=y_dtype or yt.dtype)
+        if yt.shape[0] != Xd.global_rows:
+            raise ValueError(
+                f"X and y have inconsistent lengths: {Xd.global_rows} vs "
+                f"{yt.shape[0]}")
+        return Xd, Replicated(yt, Xd.comm)
     yd = check_array(y, y_dtype, Xd.comm)
     if Xd.global_rows != yd.global_rows:
         raise ValueError(

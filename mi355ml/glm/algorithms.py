@@ -54,6 +54,13 @@ def _shard_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
 def _dist_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
                      family: str) -> Tuple[float, torch.Tensor, float]:
     """Global (loss, grad_w, grad_b) — fused kernel + one fused allreduce."""
+    from ..core.colarray import ColDistArray
+    if isinstance(X, ColDistArray):
+        # feature-parallel path: col_value_grad does its own collectives
+        from ..ops.glm import col_value_grad
+        yt = y.local if hasattr(y, "local") else y
+        loss, gw, gb = col_value_grad(X, yt, w, b, family)
+        return float(loss), gw, float(gb)
     loss, gw, gb = _shard_value_grad(X, y, w, b, family)
     buf = torch.empty(2 + gw.shape[0], dtype=torch.float64, device=gw.device)
     buf[0] = loss
@@ -66,6 +73,12 @@ def _dist_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
 
 def _local_value_grad(X: DistArray, y: DistArray, w: torch.Tensor, b: float,
                       family: str) -> Tuple[float, torch.Tensor, float]:
+    from ..core.colarray import ColDistArray
+    if isinstance(X, ColDistArray):
+        raise ValueError(
+            "admm needs per-rank row shards (local subproblems); "
+            "column-sharded X supports lbfgs/gradient_descent/"
+            "proximal_grad — SURVEY §5.7")
     loss, gw, gb = _shard_value_grad(X, y, w, b, family)
     return float(loss), gw, float(gb)
 
@@ -145,6 +158,10 @@ def newton(X: DistArray, y: DistArray, family="logistic", max_iter: int = 50,
            ) -> Tuple[torch.Tensor, float, int]:
     """Newton's method: H = XᵀDX via blocked per-shard syrk + allreduce
     (dask_glm newton; SURVEY §2.2 — driver-side lstsq on the small H)."""
+    from ..core.colarray import ColDistArray
+    if isinstance(X, ColDistArray):
+        raise ValueError("newton builds a dense [d+1,d+1] Hessian — not "
+                         "supported for column-sharded (wide) X; use lbfgs")
     fam_cls = get_family(family)
     fam = fam_cls.name
     d = X.ncols

@@ -100,8 +100,13 @@ class _GLMBase(BaseEstimator):
         w = self.coef_.to(X.device)
         b = self.intercept_
         from ..core.sparse import SparseDistArray
+        from ..core.colarray import ColDistArray, Replicated
         if isinstance(X, SparseDistArray):
             return X.dot(w).map_local(lambda z: z + b)
+        if isinstance(X, ColDistArray):
+            # feature-parallel: partial GEMV + one [m] allreduce; the
+            # replicated result row-shards on contact with metrics
+            return Replicated(X.dot(w.to(X.dtype)) + b, X.comm)
         return X.map_local(lambda t: t @ w.to(t.dtype) + b)
 
     def predict(self, X):

@@ -112,3 +112,38 @@ def glm_predict_linear(x: torch.Tensor, w: torch.Tensor, b: float
                        ) -> torch.Tensor:
     """z = X·w + b for a shard (library GEMV; not a hot op)."""
     return x @ w.to(x.dtype) + b
+
+
+def col_value_grad(X, y_full: torch.Tensor, w_full: torch.Tensor, b: float,
+                   family: str):
+    """Feature-parallel fused objective for a column-sharded X
+    (core.colarray.ColDistArray — SURVEY §5.7 wide-X mechanism).
+
+    z-pass: partial GEMV over the local column slice + ONE [m] allreduce
+    (the feature-parallel dual of the row-shard's local z). Pointwise link
+    math runs replicated (identical on every rank); the gradient slice
+    Xᵀr is LOCAL (columns already sharded) and is allgathered into the
+    full [d] vector the redundant-deterministic lbfgs driver needs.
+
+    Returns GLOBAL (loss, grad_w[d] f64, grad_b) — callers must NOT
+    allreduce again.
+    """
+    assert family in _FAMILY_ID, family
+    yd = y_full.to(device=X.device, dtype=torch.float64)
+    z = X.dot(w_full.to(X.dtype))          # replicated [m] (one allreduce)
+    zd = z.double() + float(b)
+    if family == "logistic":
+        loss = (torch.nn.functional.softplus(zd) - yd * zd).sum()
+        r = torch.sigmoid(zd) - yd
+    elif family == "normal":
+        diff = zd - yd
+        loss = 0.5 * (diff * diff).sum()
+        r = diff
+    else:  # poisson
+        ez = torch.exp(zd)
+        loss = (ez - yd * zd).sum()
+        r = ez - yd
+    gl = X.rmatvec_local(r.to(X.dtype)).double()   # [d_local], no comm
+    gw = X.gather_cols(gl)                          # full [d]
+    gb = r.sum()
+    return loss, gw, gb

@@ -187,3 +187,63 @@ def svd_compressed(X: DistArray, k: int, n_power_iter: int = 2,
         u = DistArray(Q._as2d() @ u_b[:, :k].to(X.dtype), m, X.row_offset,
                       comm)
     return u, s[:k].to(X.dtype), vh[:k].to(X.dtype)
+
+
+def svd_randomized_col(X, k: int, n_power_iter: int = 2,
+                       n_oversamples: int = 10, seed: int = 0,
+                       compute_u: bool = True):
+    """Randomized truncated SVD for a COLUMN-sharded matrix
+    (core.colarray.ColDistArray — the wide-X mechanism).
+
+    Feature-parallel duals of svd_compressed's collectives:
+      range finder   Y = Σ_r X_loc Ω_loc        → allreduce [m, ell]
+      power iter     B_loc = X_locᵀ Q (local);  Y = Σ X_loc B_loc → allreduce
+      final factor   G = Σ B_loc B_locᵀ         → allreduce [ell, ell] f64
+      eigh(G) replicated-deterministic; Vh comes out column-sharded.
+
+    Q lives replicated ([m, ell] on every rank, orthonormalized by a local
+    QR — identical on all ranks); no TSQR needed because the row axis is
+    not sharded. Returns (U [m, k] replicated tensor or None, s [k],
+    Vh_full [k, d] gathered tensor).
+    """
+    a = X.local
+    m, d = X.global_rows, X.global_cols
+    comm = X.comm
+    ell = min(k + n_oversamples, d, m)
+    # full Ω generated from the seed, each rank keeps its row slice —
+    # bit-identical results regardless of world size
+    g = mlrand.generator(seed, shard_id=0xC01, device=a.device)
+    omega = torch.randn(d, ell, generator=g, device=a.device, dtype=X.dtype)
+    om_loc = omega[X.col_offset:X.col_offset + X.local_cols]
+    Y = a @ om_loc
+    if comm.world_size > 1:
+        comm.all_reduce(Y)
+    Q, _ = torch.linalg.qr(Y)
+    for _ in range(n_power_iter):
+        b_loc = a.t() @ Q          # [d_loc, ell], no comm
+        Y = a @ b_loc
+        if comm.world_size > 1:
+            comm.all_reduce(Y)
+        Q, _ = torch.linalg.qr(Y)
+    b_loc = (Q.t() @ a).to(torch.float64)      # [ell, d_loc], no comm
+    gram = b_loc @ b_loc.t()
+    if comm.world_size > 1:
+        comm.all_reduce(gram)
+    evals, evecs = torch.linalg.eigh(gram)     # ascending
+    evals = evals.flip(0).clamp(min=0.0)
+    u_b = evecs.flip(1)                        # [ell, ell] descending
+    s = evals.sqrt()
+    safe = s.clamp(min=torch.finfo(torch.float64).tiny)
+    vh_loc = (u_b.t() @ b_loc) / safe[:, None]  # [ell, d_loc] col-sharded
+    # svd_flip parity: sign from the globally largest-|v| entry per row
+    vh_full = X.gather_cols(vh_loc.t().contiguous()).t()   # [ell, d]
+    idx = vh_full.abs().argmax(dim=1)
+    signs = torch.sign(
+        vh_full[torch.arange(vh_full.shape[0], device=vh_full.device), idx])
+    signs[signs == 0] = 1.0
+    vh_full = vh_full * signs[:, None]
+    u_b = u_b * signs[None, :]
+    u = None
+    if compute_u:
+        u = (Q.to(torch.float64) @ u_b[:, :k]).to(X.dtype)
+    return u, s[:k].to(X.dtype), vh_full[:k].to(X.dtype)
