@@ -20,6 +20,12 @@ import torch
 from ._ext import hip_ext, use_hip
 
 
+#: diagnostic: which implementation the last kmeans_assign/closest_center
+#: call took — "fused" | "chunked-fused" | "composed" (GPU tests assert the
+#: HIP path actually ran — VERDICT r1 #6)
+last_path = None
+
+
 def _fits_fused(k: int, d: int, update: bool) -> bool:
     """Mirror of kernels.hip kmeans_smem_bytes: the fused kernel stages
     centers + a 64-row x tile (+ bins) in the 160 KB LDS. Outside this
@@ -50,15 +56,28 @@ def kmeans_assign(x: torch.Tensor, centers: torch.Tensor
                 torch.zeros(k, d, dtype=torch.float64, device=x.device),
                 torch.zeros(k, dtype=torch.float64, device=x.device),
                 torch.zeros((), dtype=torch.float64, device=x.device))
+    global last_path
     if use_hip(x) and _fits_fused(k, d, update=True):
         ext = hip_ext()
         labels, sums, counts, inertia = ext.kmeans_assign(
             x.contiguous(), centers.contiguous())
+        last_path = "fused"
         return labels, sums, counts, inertia
     labels, mindist = closest_center(x, centers)
-    # deterministic centroid sums: label-sorted segment reduction per row
-    # block in fixed block order (index_add_ is atomic f64 on GPU and
-    # would break the bitwise run-to-run guarantee on this path)
+    assign_path = last_path
+    sums = _label_update(x, labels, k)
+    counts = torch.bincount(labels.long(), minlength=k).to(torch.float64)
+    inertia = mindist.to(torch.float64).sum()
+    last_path = assign_path  # the assign pass dominates; report its path
+    return labels, sums, counts, inertia
+
+
+def _label_update(x: torch.Tensor, labels: torch.Tensor, k: int
+                  ) -> torch.Tensor:
+    """Deterministic centroid sums: label-sorted segment reduction per row
+    block in fixed block order (index_add_ is atomic f64 on GPU and would
+    break the bitwise run-to-run guarantee on this path)."""
+    d = x.shape[1]
     sums = torch.zeros(k, d, dtype=torch.float64, device=x.device)
     m = x.shape[0]
     block = 4_194_304
@@ -75,9 +94,7 @@ def kmeans_assign(x: torch.Tensor, centers: torch.Tensor
         part = torch.segment_reduce(xs, "sum", offsets=offsets, axis=0,
                                     initial=0)
         sums[uniq] += part
-    counts = torch.bincount(labels.long(), minlength=k).to(torch.float64)
-    inertia = mindist.to(torch.float64).sum()
-    return labels, sums, counts, inertia
+    return sums
 
 
 def closest_center(x: torch.Tensor, centers: torch.Tensor
@@ -86,14 +103,38 @@ def closest_center(x: torch.Tensor, centers: torch.Tensor
 
     (pairwise_distances_argmin_min parity; dist clamped at 0.)
     """
+    global last_path
     assert x.dim() == 2 and centers.dim() == 2
     if x.shape[0] == 0:
         return (torch.empty(0, dtype=torch.int32, device=x.device),
                 torch.empty(0, dtype=x.dtype, device=x.device))
-    if use_hip(x) and _fits_fused(centers.shape[0], x.shape[1], update=False):
+    k, d = centers.shape
+    if use_hip(x) and _fits_fused(k, d, update=False):
         ext = hip_ext()
         labels, dists = ext.closest_center(x.contiguous(), centers.contiguous())
+        last_path = "fused"
         return labels, dists
+    if use_hip(x) and k > 256 and _fits_fused(256, d, update=False):
+        # chunked-fused (VERDICT r1 #6): the hand-written MFMA
+        # distance+argmin kernel runs per ≤256-center chunk; the running
+        # (dist, label) merge is two [m]-sized elementwise ops —
+        # negligible next to the m×d distance GEMM it rides behind.
+        ext = hip_ext()
+        xc = x.contiguous()
+        best_d = None
+        best_l = None
+        for lo in range(0, k, 256):
+            hi = min(k, lo + 256)
+            lab, dst = ext.closest_center(xc, centers[lo:hi].contiguous())
+            if best_d is None:
+                best_d, best_l = dst, lab + lo if lo else lab
+            else:
+                win = dst < best_d
+                best_d = torch.where(win, dst, best_d)
+                best_l = torch.where(win, lab + lo, best_l)
+        last_path = "chunked-fused"
+        return best_l, best_d
+    last_path = "composed"
     cn = (centers * centers).sum(dim=1)
     # blocked to bound the [m, k] temporary
     m = x.shape[0]
