@@ -266,7 +266,23 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
       float dot = 0.0f;
       if (r < rows) {
         const float* xrow = xt + r * ld;
-        for (int j = q; j < d; j += 4) dot += xrow[j] * sw[j];
+        if ((ld & 3) == 0) {
+          // b128 LDS reads: lane q covers float4 quads q, q+4, ... —
+          // 4x fewer ds_read instructions on the issue-bound z chain
+          // (r1 ablation: z+grad together exceeded the stream floor)
+          const f32x4* xr4 = (const f32x4*)xrow;
+          const f32x4* wv4 = (const f32x4*)sw;
+          const int n4 = d >> 2;
+          for (int j = q; j < n4; j += 4) {
+            const f32x4 a = xr4[j];
+            const f32x4 v = wv4[j];
+            dot += a[0] * v[0] + a[1] * v[1] + a[2] * v[2] + a[3] * v[3];
+          }
+          const int jt = (d & ~3) + q;
+          if (jt < d) dot += xrow[jt] * sw[jt];
+        } else {
+          for (int j = q; j < d; j += 4) dot += xrow[j] * sw[j];
+        }
       }
       dot += __shfl_xor(dot, 1, 4);
       dot += __shfl_xor(dot, 2, 4);
@@ -389,19 +405,24 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
   constexpr int T0 = (NTILES + 1) / 2;     // first half's tile count
   constexpr int T1 = NTILES - T0;          // second half's tile count
   const int d4 = (d + 3) & ~3;        // K padded to a multiple of 4
+  // center-row LDS stride: stride%8==4 makes the 8 half-wave b-reads of
+  // the MFMA loop (lanes lj=0..7/8..15, addr = lj*cs + lk) land on 32
+  // distinct banks; d4%8==0 (e.g. d=100 -> 104) was a 2-way conflict on
+  // every b-read (final r1 PMC: 21% bank-conflict/LDS on this kernel)
+  const int cs = d4 + (((d4 & 7) == 0) ? 4 : 0);
   const int k16 = NTILES * 16;        // centers padded to a multiple of 16
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* ct = (float*)smem_raw;                  // centers [k16 * d4]
-  float* nhc = ct + k16 * d4;                    // -0.5*‖c‖² [k16]
+  float* ct = (float*)smem_raw;                  // centers [k16 * cs]
+  float* nhc = ct + k16 * cs;                    // -0.5*‖c‖² [k16]
   float* xtbuf = nhc + k16;                      // 2 x-tile buffers
   float* xn = xtbuf + 2 * KM_TR * d4;            // row norms [KM_TR]
   int* rlab = (int*)(xn + KM_TR);                // row labels [KM_TR]
-  float* cbest = (float*)(rlab + KM_TR);         // [16][KM_TR+1] candidates
-  int* cbj = (int*)(cbest + 16 * (KM_TR + 1));   // [16][KM_TR+1]
-  float* bins = (float*)(cbj + 16 * (KM_TR + 1));  // UPDATE: [k*d]
+  float* cbest = (float*)(rlab + KM_TR);         // [16][KM_TR+4] candidates
+  int* cbj = (int*)(cbest + 16 * (KM_TR + 4));   // [16][KM_TR+4]
+  float* bins = (float*)(cbj + 16 * (KM_TR + 4));  // UPDATE: [k*d]
   float* bcnt = bins + (UPDATE ? k * d : 0);     // UPDATE: [k]
-  double* red = (double*)(smem_raw + ((((size_t)(k16 * d4 + k16 +
-      2 * KM_TR * d4 + 2 * KM_TR + 32 * (KM_TR + 1) +
+  double* red = (double*)(smem_raw + ((((size_t)(k16 * cs + k16 +
+      2 * KM_TR * d4 + 2 * KM_TR + 32 * (KM_TR + 4) +
       (UPDATE ? (k * d + k) : 0)) * 4 + 15) & ~15)));
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -412,11 +433,11 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
   const int lk = lane >> 4;    // fragment K index (0..3)
 
   // stage centers (zero-padded) + norms
-  for (int i = tid; i < k16 * d4; i += KM_NT) ct[i] = 0.0f;
+  for (int i = tid; i < k16 * cs; i += KM_NT) ct[i] = 0.0f;
   __syncthreads();
   for (int i = tid; i < k * d; i += KM_NT) {
     const int j = i / d;
-    ct[j * d4 + (i - j * d)] = cen[i];
+    ct[j * cs + (i - j * d)] = cen[i];
   }
   if (UPDATE) {
     for (int i = tid; i < k * d; i += KM_NT) bins[i] = 0.0f;
@@ -426,7 +447,7 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
   for (int j = tid; j < k16; j += KM_NT) {
     if (j < k) {
       float s = 0.0f;
-      const float* cp = ct + j * d4;
+      const float* cp = ct + j * cs;
       for (int kk = 0; kk < d4; ++kk) s += cp[kk] * cp[kk];
       nhc[j] = -0.5f * s;
     } else {
@@ -503,7 +524,7 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
       const int ri = qw * 16 + lj;
       const float* xrow = xt + ri * d4 + lk;
       const int tbase = half ? T0 : 0;
-      const float* cb = ct + (16 * tbase + lj) * d4 + lk;
+      const float* cb = ct + (16 * tbase + lj) * cs + lk;
       const int kl = d4 - 4;
       float best0 = -INFINITY, best1 = -INFINITY, best2 = -INFINITY,
             best3 = -INFINITY;
@@ -519,8 +540,8 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
     float a1 = xrow[min(4, kl)];                                            \
     float b0[MYT], b1[MYT];                                                 \
     _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                       \
-      b0[t] = cb[16 * t * d4];                                              \
-      b1[t] = cb[16 * t * d4 + min(4, kl)];                                 \
+      b0[t] = cb[16 * t * cs];                                               \
+      b1[t] = cb[16 * t * cs + min(4, kl)];                                  \
     }                                                                       \
     int k0 = 0;                                                             \
     for (; k0 + 8 <= d4; k0 += 8) {                                         \
@@ -535,8 +556,8 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
       a0 = xrow[kn0];                                                       \
       a1 = xrow[kn1];                                                       \
       _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                     \
-        b0[t] = cb[16 * t * d4 + kn0];                                      \
-        b1[t] = cb[16 * t * d4 + kn1];                                      \
+        b0[t] = cb[16 * t * cs + kn0];                                      \
+        b1[t] = cb[16 * t * cs + kn1];                                      \
       }                                                                     \
     }                                                                       \
     if (k0 < d4) {                                                          \
@@ -585,32 +606,45 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
         const int bjs[4] = {bj0, bj1, bj2, bj3};
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
-          cbest[slot * (KM_TR + 1) + rbase + g] = bests[g];
-          cbj[slot * (KM_TR + 1) + rbase + g] = bjs[g];
+          cbest[slot * (KM_TR + 4) + rbase + g] = bests[g];
+          cbj[slot * (KM_TR + 4) + rbase + g] = bjs[g];
         }
       }
       lds_barrier();
-      if (tid < KM_TR && tid < rows) {  // wave 0: one lane per row
-        const int r = tid;
-        float bv = cbest[r];
-        int bj = cbj[r];
+      {
+        // widened finalize: ALL 512 threads (8 slot-lanes per row) read 2
+        // slots each and merge with 3 shfl steps — the old 64-thread
+        // serial 16-slot scan was 0.32 ms of the 2.3 ms kernel. Slot
+        // stride KM_TR+4 (≡68 ≡ 4 mod 32 banks) keeps the (g, r)
+        // half-wave read pattern conflict-free.
+        const int r = tid >> 3;
+        const int g = tid & 7;
         const int nslots = T1 > 0 ? 16 : 8;
-#pragma unroll
-        for (int sl = 1; sl < 16; ++sl) {
-          if (sl >= nslots) break;
-          const float ov = cbest[sl * (KM_TR + 1) + r];
-          const int oj = cbj[sl * (KM_TR + 1) + r];
-          if (ov > bv || (ov == bv && oj < bj)) {
-            bv = ov;
-            bj = oj;
+        float bv = -INFINITY;
+        int bj = 0;
+        if (r < rows) {
+          bv = cbest[g * (KM_TR + 4) + r];
+          bj = cbj[g * (KM_TR + 4) + r];
+          if (nslots == 16) {
+            const float ov = cbest[(g + 8) * (KM_TR + 4) + r];
+            const int oj = cbj[(g + 8) * (KM_TR + 4) + r];
+            if (ov > bv || (ov == bv && oj < bj)) { bv = ov; bj = oj; }
           }
         }
-        labels[r0 + r] = bj;  // 64-lane coalesced store
-        if (UPDATE) {
-          rlab[r] = bj;
-          iacc += -2.0 * (double)bv;  // + Σ‖x‖² from the update pass
-        } else {
-          dists[r0 + r] = fmaxf(xn[r] - 2.0f * bv, 0.0f);
+#pragma unroll
+        for (int s = 1; s < 8; s <<= 1) {
+          const float ov = __shfl_xor(bv, s, 8);
+          const int oj = __shfl_xor(bj, s, 8);
+          if (ov > bv || (ov == bv && oj < bj)) { bv = ov; bj = oj; }
+        }
+        if (r < rows && g == 0) {
+          labels[r0 + r] = bj;
+          if (UPDATE) {
+            rlab[r] = bj;
+            iacc += -2.0 * (double)bv;  // + Σ‖x‖² from the update pass
+          } else {
+            dists[r0 + r] = fmaxf(xn[r] - 2.0f * bv, 0.0f);
+          }
         }
       }
       }  // PHASES & 8 (epilogue)
@@ -713,9 +747,10 @@ extern "C" __global__ __launch_bounds__(NT) void affine_transform_k(
 // ---------------------------------------------------------------------------
 static size_t kmeans_smem_bytes_impl(int k, int d, int update) {
   const int d4 = (d + 3) & ~3;
+  const int cs = d4 + (((d4 & 7) == 0) ? 4 : 0);  // mirrors the kernel
   const int k16 = (k + 15) & ~15;
-  size_t f32s = (size_t)k16 * d4 + k16 + (size_t)2 * 64 * d4 + 2 * 64 +
-                32 * 65;
+  size_t f32s = (size_t)k16 * cs + k16 + (size_t)2 * 64 * d4 + 2 * 64 +
+                32 * 68;
   if (update) f32s += (size_t)k * d + k;
   return (((f32s * 4 + 15) & ~15)) + 512 * sizeof(double);
 }

@@ -35,8 +35,9 @@ def _fits_fused(k: int, d: int, update: bool) -> bool:
     if k > 256:
         return False
     d4 = (d + 3) & ~3
+    cs = d4 + (4 if (d4 & 7) == 0 else 0)  # bank-conflict-free c stride
     k16 = (k + 15) & ~15
-    f32s = k16 * d4 + k16 + 2 * 64 * d4 + 2 * 64 + 32 * 65
+    f32s = k16 * cs + k16 + 2 * 64 * d4 + 2 * 64 + 32 * 68
     if update:
         f32s += k * d + k
     return f32s * 4 + 16 + 512 * 8 <= 160 * 1024 - 1024
