@@ -272,8 +272,9 @@ class Normalizer(BaseEstimator, TransformerMixin):
 
 
 class PolynomialFeatures(BaseEstimator, TransformerMixin):
-    """Degree-2/3 polynomial feature expansion, blockwise on device
-    (the reference maps sklearn's transformer per block)."""
+    """Arbitrary-degree polynomial feature expansion (+interaction_only),
+    blockwise on device (the reference maps sklearn's transformer per
+    block); column order matches sklearn's combinations ordering."""
 
     def __init__(self, degree: int = 2, interaction_only: bool = False,
                  include_bias: bool = True):
@@ -296,11 +297,15 @@ class PolynomialFeatures(BaseEstimator, TransformerMixin):
     def fit(self, X, y=None):
         X = check_array(X)
         d = X.ncols
-        if self.degree > 3:
-            raise ValueError("degree <= 3 supported")
+        if self.degree < 0:
+            raise ValueError("degree must be >= 0")
         self._combos_ = self._combos(d)
         self.n_features_in_ = d
         self.n_output_features_ = len(self._combos_)
+        if self.n_output_features_ > 100_000:
+            raise ValueError(
+                f"degree={self.degree} over {d} features expands to "
+                f"{self.n_output_features_} columns; refusing (>1e5)")
         return self
 
     def transform(self, X):

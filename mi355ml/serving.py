@@ -33,13 +33,93 @@ def _to_plain(res) -> List:
     return np.asarray(res).tolist()
 
 
+class MicroBatcher:
+    """Dynamic request batching: concurrent requests for
+    the same method are coalesced into ONE model call — rows are
+    concatenated up to max_batch (waiting up to max_wait_ms for
+    stragglers), the model runs once in a thread-pool executor, and each
+    caller gets its row slice back. One GPU pass per micro-batch instead
+    of one per request."""
+
+    def __init__(self, model: Any, max_batch: int = 65536,
+                 max_wait_ms: float = 2.0):
+        self.model = model
+        self.max_batch = max_batch
+        self.max_wait = max_wait_ms / 1e3
+        self.queue = None  # created lazily on the serving event loop
+        self._task = None
+        self.calls = 0  # model invocations (observability/tests)
+
+    def _ensure(self):
+        import asyncio
+        if self.queue is None:
+            self.queue = asyncio.Queue()
+            self._task = asyncio.get_event_loop().create_task(self._worker())
+
+    async def submit(self, name: str, x: np.ndarray):
+        import asyncio
+        self._ensure()
+        fut = asyncio.get_event_loop().create_future()
+        await self.queue.put((name, x, fut))
+        return await fut
+
+    async def _worker(self):
+        import asyncio
+        loop = asyncio.get_event_loop()
+        while True:
+            name, x, fut = await self.queue.get()
+            batch = [(x, fut)]
+            rows = x.shape[0]
+            other = []
+            deadline = loop.time() + self.max_wait
+            while rows < self.max_batch:
+                timeout = deadline - loop.time()
+                if timeout <= 0 and self.queue.empty():
+                    break
+                try:
+                    item = await asyncio.wait_for(self.queue.get(),
+                                                  max(timeout, 0.0))
+                except asyncio.TimeoutError:
+                    break
+                if item[0] != name or rows + item[1].shape[0] > \
+                        self.max_batch:
+                    other.append(item)  # different method / would overflow
+                    continue
+                batch.append((item[1], item[2]))
+                rows += item[1].shape[0]
+            for item in other:
+                self.queue.put_nowait(item)
+            xs = (np.concatenate([b[0] for b in batch], axis=0)
+                  if len(batch) > 1 else batch[0][0])
+
+            def run():
+                return _to_plain(getattr(self.model, name)(xs))
+
+            try:
+                self.calls += 1
+                res = await loop.run_in_executor(None, run)
+                lo = 0
+                for xb, f in batch:
+                    if not f.done():
+                        f.set_result(res[lo:lo + xb.shape[0]])
+                    lo += xb.shape[0]
+            except Exception as e:
+                for _, f in batch:
+                    if not f.done():
+                        f.set_exception(e)
+
+
 def create_app(model: Any, max_batch: int = 65536,
-               token: Optional[str] = None):
+               token: Optional[str] = None, batching: bool = False,
+               max_wait_ms: float = 2.0):
     from starlette.applications import Starlette
     from starlette.responses import JSONResponse, PlainTextResponse
     from starlette.routing import Route
 
     from .diagnostics import COUNTERS, device_memory
+
+    batcher = (MicroBatcher(model, max_batch, max_wait_ms)
+               if batching else None)
 
     async def health(request):
         return JSONResponse({"status": "ok"})
@@ -91,13 +171,16 @@ def create_app(model: Any, max_batch: int = 65536,
                                f"server limit of {max_batch}"},
                     status_code=413)
             try:
-                res = getattr(model, name)(x)
+                if batcher is not None:
+                    plain = await batcher.submit(name, x)
+                else:
+                    plain = _to_plain(getattr(model, name)(x))
             except Exception as e:  # surface model errors as 400s
                 COUNTERS.inc("serving.errors")
                 return JSONResponse({"detail": str(e)}, status_code=400)
             COUNTERS.inc("serving.requests")
             COUNTERS.inc("serving.rows", float(x.shape[0]))
-            return JSONResponse({"result": _to_plain(res)})
+            return JSONResponse({"result": plain})
 
         return endpoint
 
@@ -126,10 +209,15 @@ def main(argv: Optional[list] = None):
     ap.add_argument("--token", default=None,
                     help="require 'Authorization: Bearer <token>' on "
                          "inference endpoints")
+    ap.add_argument("--batching", action="store_true",
+                    help="coalesce concurrent requests into one model call")
+    ap.add_argument("--max-wait-ms", type=float, default=2.0,
+                    help="micro-batch straggler window (with --batching)")
     args = ap.parse_args(argv)
     model = joblib.load(args.model)
     uvicorn.run(create_app(model, max_batch=args.max_batch,
-                           token=args.token),
+                           token=args.token, batching=args.batching,
+                           max_wait_ms=args.max_wait_ms),
                 host=args.host, port=args.port)
 
 

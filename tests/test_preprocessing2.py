@@ -256,3 +256,21 @@ def test_label_encoder_categorical_fast_path():
     bad = DistList([s[:3], pd.Series(pd.Categorical(["x"]))])
     with pytest.raises(ValueError, match="mismatched"):
         LabelEncoder().fit(bad)
+
+
+def test_polynomial_features_arbitrary_degree():
+    # arbitrary degree + interaction_only vs sklearn
+    rng = np.random.default_rng(21)
+    a = rng.standard_normal((50, 3)).astype(np.float32)
+    for degree, inter in [(4, False), (5, False), (3, True), (1, False),
+                          (0, False)]:
+        pf = PolynomialFeatures(degree=degree, interaction_only=inter)
+        ours = pf.fit(from_numpy(a)).transform(from_numpy(a))
+        sk = SkPoly(degree=degree, interaction_only=inter).fit_transform(
+            a.astype(np.float64))
+        got = ours.gather().cpu().numpy()
+        assert got.shape == sk.shape, (degree, inter, got.shape, sk.shape)
+        np.testing.assert_allclose(got, sk, rtol=2e-3, atol=2e-3)
+    with pytest.raises(ValueError, match="refusing"):
+        PolynomialFeatures(degree=9).fit(
+            from_numpy(rng.standard_normal((10, 30)).astype(np.float32)))

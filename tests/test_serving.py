@@ -62,3 +62,51 @@ def test_serving_auth():
     assert r.status_code == 200
     # health stays open
     assert client.get("/health").status_code == 200
+
+
+def test_micro_batcher_coalesces_concurrent_requests():
+    """VERDICT r1 #10 serving batching: N concurrent submits for the same
+    method produce FEWER model calls than requests, and every caller gets
+    exactly its rows back."""
+    import asyncio
+    from mi355ml.serving import MicroBatcher
+
+    class SlowModel:
+        def __init__(self):
+            self.calls = 0
+
+        def predict(self, x):
+            self.calls += 1
+            import time
+            time.sleep(0.05)  # force queue build-up while the first runs
+            return np.asarray(x).sum(axis=1)
+
+    model = SlowModel()
+    mb = MicroBatcher(model, max_batch=1000, max_wait_ms=20.0)
+
+    async def run():
+        xs = [np.full((3, 2), float(i), dtype=np.float32) for i in range(6)]
+        results = await asyncio.gather(
+            *[mb.submit("predict", x) for x in xs])
+        return xs, results
+
+    xs, results = asyncio.new_event_loop().run_until_complete(run())
+    for x, r in zip(xs, results):
+        np.testing.assert_allclose(np.asarray(r), x.sum(axis=1))
+    assert model.calls < 6, f"no batching happened ({model.calls} calls)"
+    assert mb.calls == model.calls
+
+
+def test_serving_with_batching_endpoint():
+    from starlette.testclient import TestClient
+    from mi355ml.serving import create_app
+
+    class M:
+        def predict(self, x):
+            return np.asarray(x)[:, 0] * 2
+
+    app = create_app(M(), batching=True, max_wait_ms=1.0)
+    with TestClient(app) as client:
+        r = client.post("/predict", json={"data": [[1.0, 2.0], [3.0, 4.0]]})
+        assert r.status_code == 200
+        assert r.json()["result"] == [2.0, 6.0]

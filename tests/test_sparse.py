@@ -97,3 +97,41 @@ def test_sparse_multivector_dot():
     np.testing.assert_allclose(Z, M @ W, rtol=1e-12)
     with pytest.raises(NotImplementedError):
         S.sum(axis=1)
+
+
+def _sparse_rebalance_world2(rank, world, a_bytes, shape):
+    import scipy.sparse as ssp
+    import torch
+    from mi355ml.core.sparse import SparseDistArray
+    from mi355ml.parallel import comm as comm_mod
+    a = np.frombuffer(a_bytes, dtype=np.float32).reshape(shape).copy()
+    comm = comm_mod.default()
+    # deliberately UNEVEN shards: rank 0 gets 10 rows, rank 1 the rest
+    split = 10
+    lo, hi = (0, split) if rank == 0 else (split, shape[0])
+    local = ssp.csr_matrix(a[lo:hi])
+    S = SparseDistArray(torch.from_numpy(local.indptr.astype(np.int64)),
+                        torch.from_numpy(local.indices.astype(np.int64)),
+                        torch.from_numpy(local.data), shape[1],
+                        shape[0], lo, comm)
+    B = S.rebalance()
+    # balanced: row counts differ by at most 1
+    nl = comm.all_gather_object(B.local_rows)
+    assert max(nl) - min(nl) <= 1
+    assert sum(nl) == shape[0]
+    # content identical to the original global matrix
+    np.testing.assert_allclose(B.gather().toarray(), a, rtol=0, atol=0)
+    # offsets are the canonical contiguous layout
+    offs = comm.all_gather_object(B.row_offset)
+    assert offs == sorted(offs)
+    # sparse ops still work post-rebalance
+    w = np.random.default_rng(0).standard_normal(shape[1]).astype(np.float32)
+    z = B.dot(torch.from_numpy(w)).gather().cpu().numpy()
+    np.testing.assert_allclose(z, a @ w, rtol=1e-4, atol=1e-4)
+
+
+def test_sparse_rebalance_world2(mp_runner):
+    rng = np.random.default_rng(11)
+    a = (rng.uniform(size=(101, 7)) < 0.3).astype(np.float32) * \
+        rng.standard_normal((101, 7)).astype(np.float32)
+    mp_runner(_sparse_rebalance_world2, world=2, args=(a.tobytes(), a.shape))
