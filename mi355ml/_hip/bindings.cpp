@@ -24,6 +24,10 @@ void launch_kmeans_probe(const float*, const float*, long, int, int, int*,
 void launch_glm_probe(const float*, const float*, const float*, float, long,
                       int, int, double*, int, hipStream_t);
 size_t kmeans_smem_bytes(int k, int d, int update);
+void launch_kmeans_assign_split(const float*, const float*, long, int, int,
+                                int*, float*, float*, double*, int,
+                                hipStream_t);
+size_t kmeans_split_smem_bytes(int k, int d);
 }
 
 namespace {
@@ -127,6 +131,45 @@ std::vector<torch::Tensor> kmeans_assign(torch::Tensor x,
           cnt_slab.sum(0).to(torch::kFloat64), inertia_slab.sum()};
 }
 
+
+std::vector<torch::Tensor> kmeans_assign_split(torch::Tensor x,
+                                               torch::Tensor centers) {
+  check_f32_2d(x, "x");
+  check_f32_2d(centers, "centers");
+  TORCH_CHECK(x.dim() == 2 && centers.dim() == 2);
+  const long m = x.size(0);
+  const int d = (int)x.size(1);
+  const int k = (int)centers.size(0);
+  TORCH_CHECK(centers.size(1) == d);
+  TORCH_CHECK(k <= 256, "fused kmeans kernel supports k <= 256 (got ", k,
+              ")");
+  TORCH_CHECK(kmeans_split_smem_bytes(k, d) <= 160 * 1024 - 1024,
+              "k*d too large for the split kernel LDS (",
+              kmeans_split_smem_bytes(k, d), " bytes)");
+  const int nb = (int)std::max(1L, std::min(256L, (m + 255) / 256));
+  auto f32 = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
+  auto f64 = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
+  auto labels = torch::empty({m}, torch::TensorOptions()
+                                      .dtype(torch::kInt32)
+                                      .device(x.device()));
+  auto sums_slab = torch::zeros({nb, k, d}, f32);
+  auto cnt_slab = torch::zeros({nb, k}, f32);
+  auto inertia_slab = torch::zeros({nb}, f64);
+  launch_kmeans_assign_split(x.data_ptr<float>(), centers.data_ptr<float>(),
+                             m, d, k, labels.data_ptr<int>(),
+                             sums_slab.data_ptr<float>(),
+                             cnt_slab.data_ptr<float>(),
+                             inertia_slab.data_ptr<double>(), nb,
+                             cur_stream());
+  return {labels, sums_slab.sum(0).to(torch::kFloat64),
+          cnt_slab.sum(0).to(torch::kFloat64), inertia_slab.sum()};
+}
+
+bool kmeans_split_fits(long k, long d) {
+  return k <= 256 &&
+         kmeans_split_smem_bytes((int)k, (int)d) <= 160 * 1024 - 1024;
+}
+
 std::vector<torch::Tensor> closest_center(torch::Tensor x,
                                           torch::Tensor centers) {
   check_f32_2d(x, "x");
@@ -213,6 +256,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused GLM (loss, grad_w, grad_b) over one shard");
   mod.def("kmeans_assign", &kmeans_assign,
           "fused KMeans labels + centroid partial sums/counts/inertia");
+  mod.def("kmeans_assign_split", &kmeans_assign_split,
+          "wave-specialized assign+update (2 update waves, software "
+          "pipelined over tiles)");
+  mod.def("kmeans_split_fits", &kmeans_split_fits,
+          "does (k, d) fit the split kernel LDS budget");
   mod.def("closest_center", &closest_center,
           "fused argmin-distance labels + squared distances");
   mod.def("affine_transform", &affine_transform,

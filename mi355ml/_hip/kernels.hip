@@ -710,6 +710,314 @@ __global__ __launch_bounds__(KM_NT) void kmeans_assign_k(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Wave-specialized kmeans assign+update (UPDATE=1 only): 640 threads =
+// 8 MFMA waves (the exact flow of kmeans_assign_k) + 2 UPDATE waves that
+// run the centroid-bin accumulation for the PREVIOUS tile concurrently,
+// at full lane occupancy, reading x straight from L2 (the LDS tile of
+// t-1 is being overwritten by the t+1 prefetch, so the update re-reads
+// global memory — ~26 KB/tile, L2-hot). The old in-line update executed
+// a rows x ops instruction stream on ALL 8 waves with <=13/64 active
+// lanes; here the same stream runs on 2 waves while the MFMA waves crunch
+// the next tile. Bin sum order per column is rows-ascending, exactly the
+// old order -> bitwise-identical results.
+// ---------------------------------------------------------------------------
+#define KMS_NT 640
+
+template <int NTILES, int PHASES = 15>
+__global__ __launch_bounds__(KMS_NT) void kmeans_assign_split_k(
+    const float* __restrict__ x, const float* __restrict__ cen, long m,
+    int d, int k, int* __restrict__ labels,
+    float* __restrict__ sums_slab, float* __restrict__ cnt_slab,
+    double* __restrict__ inertia_slab) {
+  constexpr int T0 = (NTILES + 1) / 2;
+  constexpr int T1 = NTILES - T0;
+  const int d4 = (d + 3) & ~3;
+  const int cs = d4 + (((d4 & 7) == 0) ? 4 : 0);
+  const int k16 = NTILES * 16;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* ct = (float*)smem_raw;                  // centers [k16 * cs]
+  float* nhc = ct + k16 * cs;                    // -0.5*c^2 [k16]
+  float* xtbuf = nhc + k16;                      // 2 x-tile buffers
+  int* rlab2 = (int*)(xtbuf + 2 * KM_TR * d4);   // [2][KM_TR] label bufs
+  float* cbest = (float*)(rlab2 + 2 * KM_TR);    // [16][KM_TR+4]
+  int* cbj = (int*)(cbest + 16 * (KM_TR + 4));   // [16][KM_TR+4]
+  float* bins = (float*)(cbj + 16 * (KM_TR + 4));  // [k*d]
+  float* bcnt = bins + k * d;                    // [k]
+  double* red = (double*)(smem_raw + ((((size_t)(k16 * cs + k16 +
+      2 * KM_TR * d4 + 2 * KM_TR + 32 * (KM_TR + 4) + k * d + k) * 4 +
+      15) & ~15)));                              // [KMS_NT]
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;       // 0..9
+  const bool uwave = wave >= 8;    // update-specialized waves
+  const int uw = wave - 8;         // 0,1
+  const int half = wave >> 2;      // MFMA tile split (waves 0..7)
+  const int qw = wave & 3;
+  const int lj = lane & 15;
+  const int lk = lane >> 4;
+
+  for (int i = tid; i < k16 * cs; i += KMS_NT) ct[i] = 0.0f;
+  __syncthreads();
+  for (int i = tid; i < k * d; i += KMS_NT) {
+    const int j = i / d;
+    ct[j * cs + (i - j * d)] = cen[i];
+  }
+  for (int i = tid; i < k * d; i += KMS_NT) bins[i] = 0.0f;
+  for (int i = tid; i < k; i += KMS_NT) bcnt[i] = 0.0f;
+  __syncthreads();
+  for (int j = tid; j < k16; j += KMS_NT) {
+    if (j < k) {
+      float s = 0.0f;
+      const float* cp = ct + j * cs;
+      for (int kk = 0; kk < d4; ++kk) s += cp[kk] * cp[kk];
+      nhc[j] = -0.5f * s;
+    } else {
+      nhc[j] = -INFINITY;
+    }
+  }
+  __syncthreads();
+
+  double iacc = 0.0;
+  const long tiles = (m + KM_TR - 1) / KM_TR;
+  const bool use_glds = (d == d4) && ((PHASES & 1) != 0);
+  int cur = 0;
+  int par = 0;
+  long tile = blockIdx.x;
+
+  auto stage_direct = [&](float* dst, long t) {
+    // compute waves only: 512-thread stride
+    const long dr0 = t * KM_TR;
+    const int drows = (int)min((long)KM_TR, m - dr0);
+    if (d == d4) {
+      const long total4 = ((long)drows * d) >> 2;
+      const float4* src4 = (const float4*)(x + dr0 * (long)d);
+      float4* dst4 = (float4*)dst;
+      for (long i = tid; i < total4; i += KM_NT) dst4[i] = src4[i];
+    } else {
+      for (int i = tid; i < drows * d; i += KM_NT) {
+        const int rr = i / d;
+        dst[rr * d4 + (i - rr * d)] = x[dr0 * (long)d + i];
+      }
+      for (int i = tid; i < drows * (d4 - d); i += KM_NT) {
+        const int rr = i / (d4 - d);
+        dst[rr * d4 + d + (i - rr * (d4 - d))] = 0.0f;
+      }
+    }
+  };
+  auto glds8 = [&](float* dst, long t) {
+    glds_tile_w(x + t * KM_TR * (long)d, dst, KM_TR * d4, wave, lane, 8);
+  };
+
+  if (!uwave && (PHASES & 1) && tile < tiles) {
+    if (use_glds && m - tile * KM_TR >= KM_TR) {
+      glds8(xtbuf, tile);
+      vm_drain();
+    } else {
+      stage_direct(xtbuf, tile);
+    }
+  }
+  lds_barrier();
+  const long tend = tiles + gridDim.x;  // one extra drain iteration
+  for (; tile < tend; tile += gridDim.x) {
+    const bool live = tile < tiles;
+    const long r0 = tile * KM_TR;
+    const int rows = live ? (int)min((long)KM_TR, m - r0) : 0;
+    const long ptile = tile - (long)gridDim.x;  // update target
+    const bool ulive = (PHASES & 4) && tile != (long)blockIdx.x &&
+                       ptile < tiles;
+    const int prows = ulive ? (int)min((long)KM_TR, m - ptile * KM_TR) : 0;
+    float* xt = xtbuf + cur * KM_TR * d4;
+    const long ntile = tile + gridDim.x;
+    const bool pf = live && use_glds && ntile < tiles &&
+                    (m - ntile * KM_TR) >= KM_TR;
+    int* rl_cur = rlab2 + par * KM_TR;
+    const int* rl_prev = rlab2 + (par ^ 1) * KM_TR;
+    const float* xprev = x + (ulive ? ptile * (long)KM_TR * d : 0);
+
+#define KMS_UPD_SEG(RA, RB)                                                 \
+  for (int c = uw * 64 + lane; c < d; c += 128) {                           \
+    float sx2 = 0.0f;                                                       \
+    const float* xp = xprev + c;                                            \
+    int r = (RA);                                                           \
+    for (; r + 4 <= (RB); r += 4) {                                         \
+      const int l0 = min(max(rl_prev[r], 0), k - 1);                        \
+      const int l1 = min(max(rl_prev[r + 1], 0), k - 1);                    \
+      const int l2 = min(max(rl_prev[r + 2], 0), k - 1);                    \
+      const int l3 = min(max(rl_prev[r + 3], 0), k - 1);                    \
+      const float v0 = xp[(long)r * d];                                     \
+      const float v1 = xp[(long)(r + 1) * d];                               \
+      const float v2 = xp[(long)(r + 2) * d];                               \
+      const float v3 = xp[(long)(r + 3) * d];                               \
+      bins[l0 * d + c] += v0;                                               \
+      bins[l1 * d + c] += v1;                                               \
+      bins[l2 * d + c] += v2;                                               \
+      bins[l3 * d + c] += v3;                                               \
+      sx2 += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;                         \
+    }                                                                       \
+    for (; r < (RB); ++r) {                                                 \
+      const int lr = min(max(rl_prev[r], 0), k - 1);                        \
+      const float v = xp[(long)r * d];                                      \
+      bins[lr * d + c] += v;                                                \
+      sx2 += v * v;                                                         \
+    }                                                                       \
+    iacc += (double)sx2;                                                    \
+  }
+
+    if (!uwave) {
+      if (pf) glds8(xtbuf + (cur ^ 1) * KM_TR * d4, ntile);
+      if (live && (PHASES & 2)) {
+        const int ri = qw * 16 + lj;
+        const float* xrow = xt + ri * d4 + lk;
+        const int tbase = half ? T0 : 0;
+        const float* cb = ct + (16 * tbase + lj) * cs + lk;
+        const int kl = d4 - 4;
+        float best0 = -INFINITY, best1 = -INFINITY, best2 = -INFINITY,
+              best3 = -INFINITY;
+        int bj0 = 0, bj1 = 0, bj2 = 0, bj3 = 0;
+#define KM_MFMA_HALF(MYT)                                                   \
+  {                                                                         \
+    f32x4 acc[MYT];                                                         \
+    _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                       \
+      const float v = nhc[16 * (tbase + t) + lj];                           \
+      acc[t] = (f32x4){v, v, v, v};                                         \
+    }                                                                       \
+    float a0 = xrow[0];                                                     \
+    float a1 = xrow[min(4, kl)];                                            \
+    float b0[MYT], b1[MYT];                                                 \
+    _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                       \
+      b0[t] = cb[16 * t * cs];                                               \
+      b1[t] = cb[16 * t * cs + min(4, kl)];                                  \
+    }                                                                       \
+    int k0 = 0;                                                             \
+    for (; k0 + 8 <= d4; k0 += 8) {                                         \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t)                       \
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],    \
+                                                      0, 0, 0);             \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t)                       \
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1[t], acc[t],    \
+                                                      0, 0, 0);             \
+      const int kn0 = min(k0 + 8, kl);                                      \
+      const int kn1 = min(k0 + 12, kl);                                     \
+      a0 = xrow[kn0];                                                       \
+      a1 = xrow[kn1];                                                       \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                     \
+        b0[t] = cb[16 * t * cs + kn0];                                      \
+        b1[t] = cb[16 * t * cs + kn1];                                      \
+      }                                                                     \
+    }                                                                       \
+    if (k0 < d4) {                                                          \
+      _Pragma("unroll") for (int t = 0; t < MYT; ++t)                       \
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t], acc[t],    \
+                                                      0, 0, 0);             \
+    }                                                                       \
+    _Pragma("unroll") for (int t = 0; t < MYT; ++t) {                       \
+      const int j = 16 * (tbase + t) + lj;                                  \
+      if (acc[t][0] > best0) { best0 = acc[t][0]; bj0 = j; }                \
+      if (acc[t][1] > best1) { best1 = acc[t][1]; bj1 = j; }                \
+      if (acc[t][2] > best2) { best2 = acc[t][2]; bj2 = j; }                \
+      if (acc[t][3] > best3) { best3 = acc[t][3]; bj3 = j; }                \
+    }                                                                       \
+  }
+        constexpr int T1C = T1 > 0 ? T1 : 1;
+        if (half == 0) {
+          KM_MFMA_HALF(T0)
+        } else if (T1 > 0) {
+          KM_MFMA_HALF(T1C)
+        }
+#undef KM_MFMA_HALF
+#define KM_RED1(BEST, BJ)                                                  \
+  {                                                                        \
+    const float ov = __shfl_xor(BEST, 8, 16);                              \
+    const int oj = __shfl_xor(BJ, 8, 16);                                  \
+    if (ov > BEST || (ov == BEST && oj < BJ)) { BEST = ov; BJ = oj; }      \
+  }
+        KM_RED1(best0, bj0)
+        KM_RED1(best1, bj1)
+        KM_RED1(best2, bj2)
+        KM_RED1(best3, bj3)
+#undef KM_RED1
+        if (lj < 8) {
+          const int slot = half * 8 + lj;
+          const int rbase = qw * 16 + lk * 4;
+          const float bests[4] = {best0, best1, best2, best3};
+          const int bjs[4] = {bj0, bj1, bj2, bj3};
+#pragma unroll
+          for (int g = 0; g < 4; ++g) {
+            cbest[slot * (KM_TR + 4) + rbase + g] = bests[g];
+            cbj[slot * (KM_TR + 4) + rbase + g] = bjs[g];
+          }
+        }
+      }
+    } else if (ulive) {
+      // segment A of the previous tile's update rides under the LONG
+      // MFMA phase (48 of 64 rows); the short tail B rides under the
+      // finalize. An even 32/32 split exposed half of B.
+      __builtin_amdgcn_s_setprio(0);  // yield issue slots to MFMA waves
+      KMS_UPD_SEG(0, min(48, prows))
+      __builtin_amdgcn_s_setprio(1);
+    }
+    lds_barrier();  // publish visible / segment A retired
+    if (!uwave) {
+      if (live && (PHASES & 8)) {
+        const int r = tid >> 3;
+        const int g8 = tid & 7;
+        const int nslots = T1 > 0 ? 16 : 8;
+        float bv = -INFINITY;
+        int bj = 0;
+        if (r < rows) {
+          bv = cbest[g8 * (KM_TR + 4) + r];
+          bj = cbj[g8 * (KM_TR + 4) + r];
+          if (nslots == 16) {
+            const float ov = cbest[(g8 + 8) * (KM_TR + 4) + r];
+            const int oj = cbj[(g8 + 8) * (KM_TR + 4) + r];
+            if (ov > bv || (ov == bv && oj < bj)) { bv = ov; bj = oj; }
+          }
+        }
+#pragma unroll
+        for (int s = 1; s < 8; s <<= 1) {
+          const float ov = __shfl_xor(bv, s, 8);
+          const int oj = __shfl_xor(bj, s, 8);
+          if (ov > bv || (ov == bv && oj < bj)) { bv = ov; bj = oj; }
+        }
+        if (r < rows && g8 == 0) {
+          labels[r0 + r] = bj;
+          rl_cur[r] = bj;
+          iacc += -2.0 * (double)bv;
+        }
+      }
+      if (pf) {
+        vm_drain();
+      } else if (live && (PHASES & 1) && ntile < tiles) {
+        stage_direct(xtbuf + (cur ^ 1) * KM_TR * d4, ntile);
+      }
+    } else if (ulive) {
+      // segment B + the count pass while the MFMA waves finalize
+      KMS_UPD_SEG(min(48, prows), prows)
+      if (uw == 0 && lane < prows)
+        atomicAdd(&bcnt[min(max(rl_prev[lane], 0), k - 1)], 1.0f);
+    }
+#undef KMS_UPD_SEG
+    lds_barrier();  // rl_cur complete; x buffers swap
+    cur ^= 1;
+    par ^= 1;
+  }
+
+  __syncthreads();
+  for (int i = tid; i < k * d; i += KMS_NT)
+    sums_slab[(long)blockIdx.x * k * d + i] = bins[i];
+  for (int i = tid; i < k; i += KMS_NT)
+    cnt_slab[(long)blockIdx.x * k + i] = bcnt[i];
+  red[tid] = iacc;
+  __syncthreads();
+  for (int s = 512; s > 0; s >>= 1) {  // 640 is not a power of two
+    if (tid < s && tid + s < KMS_NT) red[tid] += red[tid + s];
+    __syncthreads();
+  }
+  if (tid == 0) inertia_slab[blockIdx.x] = red[0];
+}
+
 // ---------------------------------------------------------------------------
 // K9: per-column affine transform out = x*a[col] + b[col], one fused pass.
 // float4-vectorized when the row length is a multiple of 4.
@@ -753,6 +1061,38 @@ static size_t kmeans_smem_bytes_impl(int k, int d, int update) {
                 32 * 68;
   if (update) f32s += (size_t)k * d + k;
   return (((f32s * 4 + 15) & ~15)) + 512 * sizeof(double);
+}
+
+
+static size_t kmeans_split_smem_bytes_impl(int k, int d) {
+  const int d4 = (d + 3) & ~3;
+  const int cs = d4 + (((d4 & 7) == 0) ? 4 : 0);
+  const int k16 = (k + 15) & ~15;
+  size_t f32s = (size_t)k16 * cs + k16 + (size_t)2 * 64 * d4 + 2 * 64 +
+                32 * 68 + (size_t)k * d + k;
+  return (((f32s * 4 + 15) & ~15)) + 640 * sizeof(double);
+}
+
+static void kmeans_split_dispatch(const float* x, const float* cen, long m,
+                                  int d, int k, int* labels,
+                                  float* sums_slab, float* cnt_slab,
+                                  double* inertia_slab, int nblocks,
+                                  hipStream_t stream) {
+  const size_t smem = kmeans_split_smem_bytes_impl(k, d);
+  const int ntiles = ((k + 15) & ~15) / 16;
+#define KMS_LAUNCH(NTL)                                                     \
+  case NTL:                                                                 \
+    hipLaunchKernelGGL((kmeans_assign_split_k<NTL>), dim3(nblocks),         \
+                       dim3(KMS_NT), smem, stream, x, cen, m, d, k,         \
+                       labels, sums_slab, cnt_slab, inertia_slab);          \
+    break
+  switch (ntiles) {
+    KMS_LAUNCH(1); KMS_LAUNCH(2); KMS_LAUNCH(3); KMS_LAUNCH(4);
+    KMS_LAUNCH(5); KMS_LAUNCH(6); KMS_LAUNCH(7); KMS_LAUNCH(8);
+    KMS_LAUNCH(9); KMS_LAUNCH(10); KMS_LAUNCH(11); KMS_LAUNCH(12);
+    KMS_LAUNCH(13); KMS_LAUNCH(14); KMS_LAUNCH(15); KMS_LAUNCH(16);
+  }
+#undef KMS_LAUNCH
 }
 
 template <int UPDATE>
@@ -826,6 +1166,19 @@ void launch_glm_value_grad(const float* x, const float* y, const float* w,
 
 size_t kmeans_smem_bytes(int k, int d, int update) {
   return kmeans_smem_bytes_impl(k, d, update);
+}
+
+
+void launch_kmeans_assign_split(const float* x, const float* cen, long m,
+                                int d, int k, int* labels, float* sums_slab,
+                                float* cnt_slab, double* inertia_slab,
+                                int nblocks, hipStream_t stream) {
+  kmeans_split_dispatch(x, cen, m, d, k, labels, sums_slab, cnt_slab,
+                        inertia_slab, nblocks, stream);
+}
+
+size_t kmeans_split_smem_bytes(int k, int d) {
+  return kmeans_split_smem_bytes_impl(k, d);
 }
 
 void launch_kmeans_assign(const float* x, const float* cen, long m, int d,

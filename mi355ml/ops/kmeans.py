@@ -59,7 +59,17 @@ def kmeans_assign(x: torch.Tensor, centers: torch.Tensor
                 torch.zeros((), dtype=torch.float64, device=x.device))
     global last_path
     if use_hip(x) and _fits_fused(k, d, update=True):
+        import os
         ext = hip_ext()
+        # wave-specialized variant: 2 dedicated update waves pipelined one
+        # tile behind the 8 MFMA waves (bitwise-identical results);
+        # opt-out with MI355ML_KMEANS_SPLIT=0
+        if (os.environ.get("MI355ML_KMEANS_SPLIT", "1") != "0"
+                and ext.kmeans_split_fits(k, d)):
+            labels, sums, counts, inertia = ext.kmeans_assign_split(
+                x.contiguous(), centers.contiguous())
+            last_path = "fused-split"
+            return labels, sums, counts, inertia
         labels, sums, counts, inertia = ext.kmeans_assign(
             x.contiguous(), centers.contiguous())
         last_path = "fused"
