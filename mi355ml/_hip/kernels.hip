@@ -15,6 +15,7 @@
 
 #include <hip/hip_runtime.h>
 #include <math.h>
+#include <stdlib.h>
 
 #define NT 256      // threads per block
 #define GLM_TR 64     // GLM row-tile height
@@ -169,8 +170,8 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 // LDS carve (all dynamic, 16B-aligned base): w[d] | xtile[TR*ld] | y[TR] |
 // rres[TR] | red[2*NT doubles].
 // ---------------------------------------------------------------------------
-template <int FAM, int PHASES = 7>
-__global__ __launch_bounds__(NT) void glm_value_grad_k(
+template <int FAM, int PHASES = 7, int GNT = NT>
+__global__ __launch_bounds__(GNT) void glm_value_grad_k(
     const float* __restrict__ x, const float* __restrict__ y,
     const float* __restrict__ w, float b, long m, int d, int ld, int nbuf,
     double* __restrict__ slab) {  // [nblocks, d+2]: gw | loss | gb
@@ -185,16 +186,19 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
   // dedicated allocation (launcher mirrors this condition exactly).
   const bool red_in_xtbuf =
       (size_t)nbuf * GLM_TR * ld * sizeof(float) >=
-      (4 * (size_t)d + NT) * sizeof(double);
+      ((GNT / 64) * (size_t)d + GNT) * sizeof(double);
   double* red = red_in_xtbuf
-      ? (double*)xtbuf + 4 * d
+      ? (double*)xtbuf + (GNT / 64) * d
       : (double*)(smem_raw +
             (((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
               sizeof(float) + 15) & ~15));
+  constexpr int GW = GNT / 64;        // waves per block (4 or 8)
+  constexpr int RPW = GLM_TR / GW;    // rows owned per wave (16 or 8)
+  constexpr int LPR = 64 / RPW;       // z-phase lanes per row (4 or 8)
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  for (int i = tid; i < d; i += NT) sw[i] = w[i];
+  for (int i = tid; i < d; i += GNT) sw[i] = w[i];
 
   double lacc = 0.0, gbacc = 0.0;
   double gacc[GLM_CSLOTS] = {};  // per-wave column slots lane + 64*sl
@@ -215,18 +219,18 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
       const long total4 = ((long)drows * d) >> 2;
       const float4* src4 = (const float4*)(x + dr0 * (long)d);
       float4* dst4 = (float4*)xdst;
-      for (long i = tid; i < total4; i += NT) dst4[i] = src4[i];
+      for (long i = tid; i < total4; i += GNT) dst4[i] = src4[i];
     } else {
-      for (int i = tid; i < drows * d; i += NT) {
+      for (int i = tid; i < drows * d; i += GNT) {
         const int rr = i / d;
         xdst[rr * ld + (i - rr * d)] = x[dr0 * (long)d + i];
       }
     }
-    for (int i = tid; i < drows; i += NT) ydst[i] = y[dr0 + i];
+    for (int i = tid; i < drows; i += GNT) ydst[i] = y[dr0 + i];
   };
 
   auto stage_glds = [&](float* xdst, float* ydst, long t) {
-    glds_tile(x + t * GLM_TR * (long)d, xdst, GLM_TR * d, wave, lane);
+    glds_tile_w(x + t * GLM_TR * (long)d, xdst, GLM_TR * d, wave, lane, GW);
     if (wave == 0) {  // y tile: one 4-byte-per-lane DMA (64 floats)
       auto gsrc = (const __attribute__((address_space(1))) char*)
           (const void*)(y + t * GLM_TR);
@@ -259,10 +263,10 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
       stage_glds(xtbuf + nxt * GLM_TR * ld, ytbuf + nxt * GLM_TR, ntile);
     // z-phase: WAVE-LOCAL (wave w owns rows [16w, 16w+16)) so the z->grad
     // dependency never crosses waves and the intra-tile barrier disappears
-    const int wrow0 = wave * 16;
+    const int wrow0 = wave * RPW;
     if (PHASES & 2) {
-      const int r = wrow0 + (lane >> 2);
-      const int q = lane & 3;
+      const int r = wrow0 + lane / LPR;
+      const int q = lane % LPR;
       float dot = 0.0f;
       if (r < rows) {
         const float* xrow = xt + r * ld;
@@ -273,7 +277,7 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
           const f32x4* xr4 = (const f32x4*)xrow;
           const f32x4* wv4 = (const f32x4*)sw;
           const int n4 = d >> 2;
-          for (int j = q; j < n4; j += 4) {
+          for (int j = q; j < n4; j += LPR) {
             const f32x4 a = xr4[j];
             const f32x4 v = wv4[j];
             dot += a[0] * v[0] + a[1] * v[1] + a[2] * v[2] + a[3] * v[3];
@@ -281,11 +285,11 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
           const int jt = (d & ~3) + q;
           if (jt < d) dot += xrow[jt] * sw[jt];
         } else {
-          for (int j = q; j < d; j += 4) dot += xrow[j] * sw[j];
+          for (int j = q; j < d; j += LPR) dot += xrow[j] * sw[j];
         }
       }
-      dot += __shfl_xor(dot, 1, 4);
-      dot += __shfl_xor(dot, 2, 4);
+#pragma unroll
+      for (int s = 1; s < LPR; s <<= 1) dot += __shfl_xor(dot, s, LPR);
       if (r < rows && q == 0) {
         double li, ri;
         if (FAM == 0) {
@@ -321,7 +325,7 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
     // waves once at kernel end). No barrier between z and grad: the LDS
     // dependency is within the wave.
     if (PHASES & 4) {
-      const int rend = min(16, rows - wrow0);
+      const int rend = min(RPW, rows - wrow0);
 #pragma unroll
       for (int sl = 0; sl < GLM_CSLOTS; ++sl) {
         const int c = lane + 64 * sl;
@@ -346,7 +350,7 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
   // array cost 2 KB of LDS and the WG/CU occupancy step at d~100)
   red[tid] = lacc;
   __syncthreads();
-  for (int s = NT / 2; s > 0; s >>= 1) {
+  for (int s = GNT / 2; s > 0; s >>= 1) {
     if (tid < s) red[tid] += red[tid + s];
     __syncthreads();
   }
@@ -354,22 +358,25 @@ __global__ __launch_bounds__(NT) void glm_value_grad_k(
   __syncthreads();
   red[tid] = gbacc;
   __syncthreads();
-  for (int s = NT / 2; s > 0; s >>= 1) {
+  for (int s = GNT / 2; s > 0; s >>= 1) {
     if (tid < s) red[tid] += red[tid + s];
     __syncthreads();
   }
   if (tid == 0) slab[(long)blockIdx.x * (d + 2) + d + 1] = red[0];
-  // combine the 4 waves' gradient partials through LDS (xtbuf reusable now)
-  double* gred = (double*)xtbuf;  // [4 * d] doubles (d <= 512 -> 16 KB)
+  // combine the waves' gradient partials through LDS (xtbuf reusable now)
+  double* gred = (double*)xtbuf;  // [GW * d] doubles
 #pragma unroll
   for (int sl = 0; sl < GLM_CSLOTS; ++sl) {
     const int c = lane + 64 * sl;
     if (c < d) gred[wave * d + c] = gacc[sl];
   }
   __syncthreads();
-  for (int c = tid; c < d; c += NT)
-    slab[(long)blockIdx.x * (d + 2) + c] =
-        gred[c] + gred[d + c] + gred[2 * d + c] + gred[3 * d + c];
+  for (int c = tid; c < d; c += GNT) {
+    double s = 0.0;
+#pragma unroll
+    for (int ww = 0; ww < GW; ++ww) s += gred[ww * d + c];
+    slab[(long)blockIdx.x * (d + 2) + c] = s;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1139,29 +1146,42 @@ void launch_glm_value_grad(const float* x, const float* y, const float* w,
   // d%4==0: contiguous LDS image (float4 staging, <=2-way bank conflicts);
   // otherwise odd stride (conflict-free, scalar staging).
   const int ld = ((d & 3) == 0) ? d : (d | 1);
-  auto smem_for = [&](int nbuf) {
+  // 256 threads measured FASTER than 512 (10.27 vs 11.45 ms at 1e8x100:
+  // halving rows-per-wave raises per-wave fixed cost more than the extra
+  // waves hide latency); MI355ML_GLM_NT=512 keeps the falsified arm
+  // reproducible.
+  static const int gnt = [] {
+    const char* e = getenv("MI355ML_GLM_NT");
+    return (e && atoi(e) == 512) ? 512 : 256;
+  }();
+  auto smem_for = [&](int nbuf, int GNT) {
     const bool red_in_xtbuf =
         (size_t)nbuf * GLM_TR * ld * sizeof(float) >=
-        (4 * (size_t)d + NT) * sizeof(double);
+        ((GNT / 64) * (size_t)d + GNT) * sizeof(double);
     return ((((((d + 3) & ~3) + nbuf * GLM_TR * ld + (nbuf + 1) * GLM_TR) *
               sizeof(float) + 15) & ~15)) +
-           (red_in_xtbuf ? 0 : NT * sizeof(double));
+           (red_in_xtbuf ? 0 : GNT * sizeof(double));
   };
-  const int nbuf = (smem_for(2) <= 158 * 1024) ? 2 : 1;
-  const size_t smem = smem_for(nbuf);
-  switch (family) {
-    case 0:
-      hipLaunchKernelGGL((glm_value_grad_k<0>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, nbuf, slab);
-      break;
-    case 1:
-      hipLaunchKernelGGL((glm_value_grad_k<1>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, nbuf, slab);
-      break;
-    default:
-      hipLaunchKernelGGL((glm_value_grad_k<2>), dim3(nblocks), dim3(NT), smem,
-                         stream, x, y, w, b, m, d, ld, nbuf, slab);
+  const int nbuf = (smem_for(2, gnt) <= 158 * 1024) ? 2 : 1;
+  const size_t smem = smem_for(nbuf, gnt);
+#define GLM_LAUNCH(FAM, GNTV)                                               \
+  hipLaunchKernelGGL((glm_value_grad_k<FAM, 7, GNTV>), dim3(nblocks),       \
+                     dim3(GNTV), smem, stream, x, y, w, b, m, d, ld, nbuf,  \
+                     slab)
+  if (gnt == 512) {
+    switch (family) {
+      case 0: GLM_LAUNCH(0, 512); break;
+      case 1: GLM_LAUNCH(1, 512); break;
+      default: GLM_LAUNCH(2, 512);
+    }
+  } else {
+    switch (family) {
+      case 0: GLM_LAUNCH(0, 256); break;
+      case 1: GLM_LAUNCH(1, 256); break;
+      default: GLM_LAUNCH(2, 256);
+    }
   }
+#undef GLM_LAUNCH
 }
 
 size_t kmeans_smem_bytes(int k, int d, int update) {
