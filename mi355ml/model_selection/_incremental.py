@@ -109,6 +109,7 @@ class BaseIncrementalSearchCV(BaseEstimator):
         self._n_candidates = len(cands)
         trials = [_Trial(i, p) for i, p in enumerate(cands)]
         owned = {t.tid: t for t in trials if t.tid % comm.world_size == comm.rank}
+        self._owner_of = {t.tid: t.tid % comm.world_size for t in trials}
         for t in owned.values():
             t.model = clone(self.estimator).set_params(**t.params)
 
@@ -176,6 +177,29 @@ class BaseIncrementalSearchCV(BaseEstimator):
             # estimators): the winner of the LAST round is by definition a
             # trial _adapt just dropped, and its trained+scored model — not
             # a hasty refit — must become best_estimator_ (advisor r1 high).
+            #
+            # Work-stealing analog (SURVEY §2.8 / stealing.py:141 cost
+            # guard): survivors are REBALANCED across ranks by next-round
+            # budget, greedily onto the least-loaded rank. The transfer
+            # cost the reference's guard worries about is ZERO here — the
+            # stepped models already traveled to every rank for global
+            # scoring — so rebalancing is always profitable. The
+            # assignment is computed redundantly+deterministically.
+            if budgets and comm.world_size > 1:
+                loads = [0.0] * comm.world_size
+                for tid, n_pf in sorted(budgets.items(),
+                                        key=lambda kv: (-kv[1], kv[0])):
+                    r = min(range(comm.world_size),
+                            key=lambda i: (loads[i], i))
+                    loads[r] += n_pf
+                    self._owner_of[tid] = r
+                    t = next(tt for tt in trials if tt.tid == tid)
+                    if r == comm.rank:
+                        if t.tid not in owned:
+                            t.model = models[tid][1]  # adopt stolen model
+                            owned[t.tid] = t
+                    elif t.tid in owned:
+                        del owned[t.tid]  # moved away (copy was global)
 
         # best trial
         alive = [t for t in trials if t.history]
@@ -183,7 +207,7 @@ class BaseIncrementalSearchCV(BaseEstimator):
         self.best_index_ = best.tid
         self.best_params_ = best.params
         self.best_score_ = best.score
-        owner = best.tid % comm.world_size
+        owner = self._owner_of.get(best.tid, best.tid % comm.world_size)
         model = None
         if comm.rank == owner:
             model = owned.get(best.tid)

@@ -260,7 +260,8 @@ class _BonusDummy:
         return self
 
     def score(self, X, y):
-        return float((np.asarray(y) == 1).mean()) * self.bonus
+        b = self.bonus[0] if isinstance(self.bonus, list) else self.bonus
+        return float((np.asarray(y) == 1).mean()) * b
 
 
 def _skewed_scoring_world2(rank, world, n):
@@ -359,3 +360,31 @@ def test_grid_search_multimetric():
                        scoring=["accuracy", "neg_log_loss"], cv=2)
     with pytest.raises(ValueError, match="refit"):
         bad.fit(from_numpy(X), from_numpy(y))
+
+
+def _stealing_world2(rank, world, n):
+    # 8 candidates; the two best (tids 0 and 2, highest bonus) are both
+    # rank-0-pinned under tid%world. The rebalance must spread the round-2
+    # survivors across BOTH ranks (transfer is free: models already
+    # traveled for global scoring), and best_estimator_ must still be the
+    # trained winner broadcast from its CURRENT owner.
+    rng = np.random.default_rng(1)
+    X = rng.standard_normal((n, 4)).astype(np.float32)
+    y = (rng.uniform(size=n) < 0.5).astype(np.float32)
+    bonuses = [8.0, 1.0, 7.0, 2.0, 3.0, 4.0, 5.0, 6.0]
+    sha = SuccessiveHalvingSearchCV(
+        _BonusDummy(), {"bonus": [[b] for b in bonuses]},
+        n_initial_parameters="grid", n_initial_iter=1, max_iter=9,
+        random_state=0)
+    # grid over a list-valued param: each candidate gets one bonus value
+    sha.fit(from_numpy(X), from_numpy(y))
+    assert sha.best_params_["bonus"] == [8.0]
+    survivors = [tid for tid, b in enumerate(bonuses) if b >= 7.0]
+    owners = {tid: sha._owner_of[tid] for tid in survivors}
+    assert set(owners.values()) == {0, 1}, owners
+    assert isinstance(sha.best_estimator_, _BonusDummy)
+    assert sha.best_estimator_.bonus == [8.0]
+
+
+def test_trial_stealing_rebalances_world2(mp_runner):
+    mp_runner(_stealing_world2, world=2, args=(600,))
