@@ -962,7 +962,7 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split_k(
       // MFMA phase (48 of 64 rows); the short tail B rides under the
       // finalize. An even 32/32 split exposed half of B.
       __builtin_amdgcn_s_setprio(0);  // yield issue slots to MFMA waves
-      KMS_UPD_SEG(0, min(48, prows))
+      KMS_UPD_SEG(0, min(56, prows))
       __builtin_amdgcn_s_setprio(1);
     }
     lds_barrier();  // publish visible / segment A retired
@@ -1001,7 +1001,7 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split_k(
       }
     } else if (ulive) {
       // segment B + the count pass while the MFMA waves finalize
-      KMS_UPD_SEG(min(48, prows), prows)
+      KMS_UPD_SEG(min(56, prows), prows)
       if (uw == 0 && lane < prows)
         atomicAdd(&bcnt[min(max(rl_prev[lane], 0), k - 1)], 1.0f);
     }
