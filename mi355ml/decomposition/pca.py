@@ -34,6 +34,9 @@ class PCA(BaseEstimator, TransformerMixin):
         return "randomized" if k < 0.8 * d else "full"
 
     def _fit(self, X: DistArray, need_u: bool = False):
+        from ..core.colarray import ColDistArray
+        if isinstance(X, ColDistArray):
+            return self._fit_col(X, need_u)
         d = X.ncols
         n = X.global_rows
         k = self.n_components if self.n_components is not None else min(n, d)
@@ -75,6 +78,42 @@ class PCA(BaseEstimator, TransformerMixin):
         self.n_samples_ = n
         return Uk, s, Vh
 
+    def _fit_col(self, X, need_u: bool = False):
+        """Wide-X feature-parallel PCA: per-column centering
+        is LOCAL on a column shard; the SVD is the col-sharded randomized
+        range finder (ops.linalg.svd_randomized_col)."""
+        from ..core.colarray import ColDistArray, Replicated
+        from ..ops.linalg import svd_randomized_col
+        d = X.ncols
+        n = X.global_rows
+        k = self.n_components if self.n_components is not None else min(n, d)
+        mean_loc = X.local.mean(dim=0).to(X.dtype)
+        Xc = ColDistArray(X.local - mean_loc[None, :], d, X.col_offset,
+                          X.comm)
+        u, s, Vh = svd_randomized_col(
+            Xc, k, n_power_iter=max(int(self.iterated_power or 0), 2),
+            seed=int(self.random_state or 0), compute_u=need_u)
+        n_f = float(n)
+        exp_var = (s.double() ** 2) / max(n_f - 1, 1.0)
+        total_var = float(
+            Xc.var(axis=0).double().sum() * n_f / max(n_f - 1, 1.0))
+        self.mean_ = X.mean(axis=0).to(X.dtype)   # full [d] (gathered)
+        self.components_ = Vh
+        self.singular_values_ = s
+        self.explained_variance_ = exp_var.to(X.dtype)
+        self.explained_variance_ratio_ = (
+            (exp_var / max(total_var, 1e-300)).to(X.dtype))
+        if k < min(n, d):
+            self.noise_variance_ = (
+                (total_var - float(exp_var.sum())) / (min(n, d) - k))
+        else:
+            self.noise_variance_ = 0.0
+        self.n_components_ = k
+        self.n_features_in_ = d
+        self.n_samples_ = n
+        Uk = Replicated(u, X.comm) if need_u else None
+        return Uk, s, Vh
+
     def fit(self, X, y=None):
         X = check_array(X)
         self._fit(X)
@@ -94,6 +133,18 @@ class PCA(BaseEstimator, TransformerMixin):
         X = check_array(X)
         V = self.components_.to(X.device)
         mean = self.mean_.to(X.device)
+        from ..core.colarray import ColDistArray, Replicated
+        if isinstance(X, ColDistArray):
+            # feature-parallel: Σ_r (X_loc - μ_loc) V_locᵀ → one [m, k]
+            # allreduce
+            lo, hi = X.col_offset, X.col_offset + X.local_cols
+            z = (X.local - mean[lo:hi][None, :]) @ V[:, lo:hi].to(X.dtype).t()
+            if X.comm.world_size > 1:
+                X.comm.all_reduce(z)
+            if self.whiten:
+                z = z * (1.0 / self.explained_variance_.to(
+                    z.device).sqrt()).to(z.dtype)
+            return Replicated(z, X.comm)
         if self.whiten:
             inv = (1.0 / self.explained_variance_.to(X.device).sqrt())
 

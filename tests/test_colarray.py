@@ -133,3 +133,50 @@ def _col_shard_layout_world2(rank, world, n, f):
 
 def test_col_shard_layout_world2(mp_runner):
     mp_runner(_col_shard_layout_world2, world=2, args=(300, 33))
+
+
+def test_col_pca_matches_dense_world1():
+    from mi355ml.decomposition import PCA
+    rng = np.random.default_rng(7)
+    U = rng.standard_normal((500, 6)).astype(np.float32) * 3
+    V = rng.standard_normal((6, 120)).astype(np.float32)
+    X = U @ V + 0.05 * rng.standard_normal((500, 120)).astype(np.float32) \
+        + rng.standard_normal(120).astype(np.float32)[None, :]
+    k = 5
+    p = PCA(n_components=k, svd_solver="randomized", iterated_power=4,
+            random_state=0)
+    Z = p.fit_transform(from_numpy_cols(X))
+    from sklearn.decomposition import PCA as SkPCA
+    sk = SkPCA(n_components=k, svd_solver="full").fit(X.astype(np.float64))
+    np.testing.assert_allclose(p.singular_values_.numpy(),
+                               sk.singular_values_, rtol=2e-2)
+    np.testing.assert_allclose(p.mean_.numpy(), X.mean(axis=0), atol=1e-3)
+    # subspace agreement
+    M = p.components_.numpy() @ sk.components_.T
+    np.testing.assert_allclose(np.abs(np.diag(M)), np.ones(k), atol=5e-2)
+    # transform path (centering + projection + allreduce)
+    Zt = p.transform(from_numpy_cols(X)).gather().numpy()
+    ref = (X - X.mean(axis=0)) @ p.components_.numpy().T
+    np.testing.assert_allclose(Zt, ref, rtol=1e-3, atol=1e-2)
+    np.testing.assert_allclose(Z.gather().numpy(), Zt, rtol=1e-3, atol=5e-2)
+
+
+def _col_pca_world2(rank, world, X_bytes, n, f):
+    from mi355ml.decomposition import PCA
+    X = np.frombuffer(X_bytes, dtype=np.float32).reshape(n, f).copy()
+    p = PCA(n_components=4, svd_solver="randomized", iterated_power=4,
+            random_state=0)
+    p.fit(from_numpy_cols(X))
+    s_exact = np.linalg.svd(X - X.mean(axis=0),
+                            compute_uv=False)[:4]
+    np.testing.assert_allclose(p.singular_values_.numpy(), s_exact,
+                               rtol=5e-2)
+    assert p.components_.shape == (4, f)
+
+
+def test_col_pca_world2(mp_runner):
+    rng = np.random.default_rng(8)
+    U = rng.standard_normal((400, 4)).astype(np.float32) * 2
+    V = rng.standard_normal((4, 60)).astype(np.float32)
+    X = (U @ V + 0.05 * rng.standard_normal((400, 60))).astype(np.float32)
+    mp_runner(_col_pca_world2, world=2, args=(X.tobytes(), 400, 60))
