@@ -255,6 +255,18 @@ class DistArray:
         """Materialize the full array (dask .compute() parity)."""
         return self.gather()
 
+    def take(self, indices) -> "DistArray":
+        """Global fancy row selection (module-level take)."""
+        return take(self, indices)
+
+    def cumsum(self, axis: int = 0) -> "DistArray":
+        assert axis == 0, "cumsum supports axis=0"
+        return _cumop(self, "sum")
+
+    def cumprod(self, axis: int = 0) -> "DistArray":
+        assert axis == 0, "cumprod supports axis=0"
+        return _cumop(self, "prod")
+
     def rebalance(self) -> "DistArray":
         """Redistribute rows to the canonical balanced contiguous layout
         via point-to-point sends (the C8/C9 gather_dep / rechunk analog:
@@ -342,3 +354,141 @@ def from_local(local: torch.Tensor, comm: Optional[Communicator] = None
     counts = [int(c.item()) for c in comm.all_gather(n)]
     offset = sum(counts[: comm.rank])
     return DistArray(local, sum(counts), offset, comm)
+
+
+def _fetch_rows(src: DistArray, ids_per_rank) -> torch.Tensor:
+    """Return THIS rank's requested global rows of `src` as a local tensor
+    (order preserved). ids_per_rank[r] (int64 cpu tensor, identical on
+    every rank) lists the global row ids rank r wants; rows move with one
+    point-to-point send per (owner, requester) pair (the C8 gather_dep
+    analog over the direct xGMI link)."""
+    import torch.distributed as dist
+    comm = src.comm
+    W = comm.world_size
+    dev = src.device
+    my_ids = ids_per_rank[comm.rank]
+    tail = tuple(src.local.shape[1:])
+    if W == 1:
+        return src.local[my_ids.to(dev)]
+    # every rank learns every shard's [offset, len) span
+    me = torch.tensor([src.row_offset, src.local.shape[0]],
+                      dtype=torch.int64,
+                      device=dev if dev.type == "cuda" else "cpu")
+    spans = [s.cpu() for s in comm.all_gather(me)]
+    starts = torch.tensor([int(s[0]) for s in spans])
+    order = torch.argsort(starts)
+    sorted_starts = starts[order]
+    out = torch.empty((my_ids.shape[0],) + tail, dtype=src.dtype,
+                      device=dev)
+    ops = []
+    fills = []
+    for r in range(W):
+        ids = ids_per_rank[r]
+        if ids.numel() == 0:
+            continue
+        pos_in_sorted = torch.searchsorted(sorted_starts, ids,
+                                           right=True) - 1
+        owner = order[pos_in_sorted.clamp(min=0)]
+        for o in range(W):
+            sel = owner == o
+            if not bool(sel.any()):
+                continue
+            if o != comm.rank and r != comm.rank:
+                continue
+            lids = (ids[sel] - int(starts[o])).to(dev)
+            pos = torch.nonzero(sel, as_tuple=False).reshape(-1)
+            if o == comm.rank and r == comm.rank:
+                out[pos.to(dev)] = src.local[lids]
+            elif o == comm.rank:
+                ops.append(dist.P2POp(dist.isend,
+                                      src.local[lids].contiguous(), r))
+            else:  # r == comm.rank, remote owner
+                buf = torch.empty((int(sel.sum()),) + tail,
+                                  dtype=src.dtype, device=dev)
+                ops.append(dist.P2POp(dist.irecv, buf, o))
+                fills.append((pos.to(dev), buf))
+    if ops:
+        for req in dist.batch_isend_irecv(ops):
+            req.wait()
+    for pos, buf in fills:
+        out[pos] = buf
+    return out
+
+
+def _canonical_splits(n: int, W: int):
+    base, rem = divmod(n, W)
+    bounds = [0]
+    for r in range(W):
+        bounds.append(bounds[-1] + base + (1 if r < rem else 0))
+    return bounds
+
+
+def take(src: DistArray, indices) -> DistArray:
+    """Global fancy row selection src[indices] (the dask slicing `take`,
+    dask/array/slicing.py:585): every rank passes the SAME index vector;
+    the result is the canonical contiguous row-shard of the selected rows
+    in order. Duplicate and negative indices allowed."""
+    comm = src.comm
+    idx = torch.as_tensor(indices, dtype=torch.int64).cpu().reshape(-1)
+    if idx.numel() and (int(idx.min()) < -src.global_rows
+                        or int(idx.max()) >= src.global_rows):
+        raise IndexError("take index out of range")
+    idx = torch.where(idx < 0, idx + src.global_rows, idx)
+    bounds = _canonical_splits(idx.shape[0], comm.world_size)
+    ids_per_rank = [idx[bounds[r]:bounds[r + 1]]
+                    for r in range(comm.world_size)]
+    local = _fetch_rows(src, ids_per_rank)
+    return DistArray(local, idx.shape[0], bounds[comm.rank], comm)
+
+
+def concat(arrays) -> DistArray:
+    """Row-wise concatenation [A; B; ...] (dask concatenate axis=0 /
+    stack of row blocks): the result is the canonical contiguous shard of
+    the stacked row space."""
+    arrays = list(arrays)
+    assert arrays, "concat of nothing"
+    comm = arrays[0].comm
+    total = sum(a.global_rows for a in arrays)
+    bounds = _canonical_splits(total, comm.world_size)
+    parts_per_rank: list = [[] for _ in range(comm.world_size)]
+    off = 0
+    pieces = []
+    for A in arrays:
+        ids_per_rank = []
+        for r in range(comm.world_size):
+            lo = max(bounds[r], off) - off
+            hi = min(bounds[r + 1], off + A.global_rows) - off
+            ids_per_rank.append(
+                torch.arange(lo, max(hi, lo), dtype=torch.int64))
+        pieces.append(_fetch_rows(A, ids_per_rank))
+        off += A.global_rows
+    local = torch.cat(pieces, dim=0) if len(pieces) > 1 else pieces[0]
+    return DistArray(local, total, bounds[comm.rank], comm)
+
+
+def _cumop(X: DistArray, op: str) -> DistArray:
+    """Shared cumsum/cumprod machinery: per-shard inclusive scan + an
+    exclusive cross-rank prefix of shard totals (the reference's
+    cumreduction, dask/array/reductions.py:1298 — the O(log W) tree
+    collapses to one small all-gather at 8 ranks)."""
+    comm = X.comm
+    loc = X.local.to(torch.float64)
+    if op == "sum":
+        cs = torch.cumsum(loc, dim=0)
+    else:
+        cs = torch.cumprod(loc, dim=0)
+    if comm.world_size > 1 and X.global_rows:
+        if loc.shape[0]:
+            tot = cs[-1:].clone()
+        else:
+            shape = (1,) + tuple(loc.shape[1:])
+            fill = 0.0 if op == "sum" else 1.0
+            tot = torch.full(shape, fill, dtype=torch.float64,
+                             device=loc.device)
+        parts = comm.all_gather(tot.contiguous())
+        for r in range(comm.rank):
+            if op == "sum":
+                cs = cs + parts[r]
+            else:
+                cs = cs * parts[r]
+    return DistArray(cs.to(X.dtype), X.global_rows, X.row_offset, X.comm)

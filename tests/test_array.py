@@ -168,3 +168,42 @@ def _empty_rank_world2(rank, world):
 
 def test_empty_rank_world2(mp_runner):
     mp_runner(_empty_rank_world2, world=2)
+
+
+def _take_concat_cum_world2(rank, world, n, f):
+    import torch
+    from mi355ml.core.array import concat, from_numpy, take
+    rng = np.random.default_rng(17)
+    a = rng.standard_normal((n, f)).astype(np.float32)
+    b = rng.standard_normal((n // 2, f)).astype(np.float32)
+    A, B = from_numpy(a), from_numpy(b)
+    # fancy take with duplicates, negatives, cross-shard pulls
+    idx = rng.integers(-n, n, 3 * n // 2)
+    got = take(A, idx).gather().cpu().numpy()
+    np.testing.assert_array_equal(got, a[idx])
+    # result is canonically sharded
+    T = take(A, idx)
+    assert T.local.shape[0] in ((len(idx)) // world, len(idx) // world + 1)
+    # concat
+    C = concat([A, B])
+    np.testing.assert_array_equal(C.gather().cpu().numpy(),
+                                  np.vstack([a, b]))
+    assert C.global_rows == n + n // 2
+    # cumsum / cumprod across the shard boundary
+    np.testing.assert_allclose(A.cumsum().gather().cpu().numpy(),
+                               a.cumsum(axis=0), rtol=1e-5, atol=1e-4)
+    p = np.abs(a[:40]) + 0.5
+    P = from_numpy(p)
+    np.testing.assert_allclose(P.cumprod().gather().cpu().numpy(),
+                               p.cumprod(axis=0), rtol=1e-4, atol=1e-4)
+
+
+def test_take_concat_cum_world2(mp_runner):
+    mp_runner(_take_concat_cum_world2, world=2, args=(120, 3))
+
+
+def test_take_out_of_range_raises():
+    from mi355ml.core.array import from_numpy, take
+    A = from_numpy(np.zeros((10, 2), dtype=np.float32))
+    with pytest.raises(IndexError):
+        take(A, [10])
