@@ -12,7 +12,8 @@ Reference capability map: SURVEY.md (mrocklin/dask-ml lineage).
 __version__ = "0.1.0"
 
 from . import config
-from .core.array import DistArray, from_numpy, shard_rows
+from .core.array import DistArray, concat, from_numpy, shard_rows, take
+from .core.colarray import ColDistArray, from_numpy_cols, shard_cols
 from .core.bag import DistList
 from .parallel import comm
 
@@ -22,6 +23,7 @@ from . import (cluster, compose, dataframe, datasets, decomposition, diagnostics
 
 __all__ = [
     "config", "comm", "DistArray", "DistList", "from_numpy", "shard_rows",
+    "ColDistArray", "from_numpy_cols", "shard_cols", "take", "concat",
     "cluster", "compose", "dataframe", "datasets", "decomposition", "diagnostics", "ensemble",
     "feature_extraction", "impute", "linear_model", "metrics",
     "model_selection", "naive_bayes", "preprocessing", "utils", "wrappers",
