@@ -15,7 +15,13 @@ for m, d, k in [(4_000_000, 100, 100), (4_000_000, 50, 100),
     la, sa, ca, ia = ext.kmeans_assign(x, c)
     lb, sb, cb, ib = ext.kmeans_assign_split(x, c)
     ok_l = bool((la == lb).all())
-    ok_s = bool((sa == sb).all())
+    # d<=64 routes to the 128-row-tile kernel whose block->row assignment
+    # differs from the inline kernel -> f32 bin partials differ in low
+    # bits (both are run-to-run deterministic); compare with tolerance
+    if (d + 3) & ~3 <= 64 and k <= 128:
+        ok_s = bool(torch.allclose(sa, sb, rtol=1e-6, atol=1e-2))
+    else:
+        ok_s = bool((sa == sb).all())
     ok_c = bool((ca == cb).all())
     # inertia: same f64 partials, different reduction-tree order across
     # the 640- vs 512-thread red[] -> equal to ~1e-12 rel, not bitwise
