@@ -34,7 +34,7 @@ def _run_bench(nproc, extra, timeout=420):
     return json.loads(line)
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_bench_distributed_contract(world):
     d = _run_bench(world, ["--steps", "2", "--warmup", "1",
                            "--rows", "100000"])
