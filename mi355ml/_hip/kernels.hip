@@ -1356,6 +1356,223 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split128_k(
   if (tid == 0) inertia_slab[blockIdx.x] = red[0];
 }
 
+
+// ---------------------------------------------------------------------------
+// 128-row-tile closest_center (UPDATE=0): 512 threads, each wave owns 16
+// rows and accumulates ALL center tiles -> in-wave full argmax (no LDS
+// candidate publish / finalize barrier), per-tile fixed costs amortized
+// over 2x rows. No bins -> fits LDS up to d~128 at k<=256. Speeds the
+// k-means|| init rounds, predict, and the chunked k>256 path (the r1
+// TODO's "128-row UPDATE=0 tiles").
+// ---------------------------------------------------------------------------
+template <int NTILES, int PHASES = 15>
+__global__ __launch_bounds__(KM_NT) void closest_center128_k(
+    const float* __restrict__ x, const float* __restrict__ cen, long m,
+    int d, int k, int* __restrict__ labels, float* __restrict__ dists) {
+  const int d4 = (d + 3) & ~3;
+  const int cs = d4 + (((d4 & 7) == 0) ? 4 : 0);
+  const int k16 = NTILES * 16;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* ct = (float*)smem_raw;                  // centers [k16 * cs]
+  float* nhc = ct + k16 * cs;                    // -0.5*c^2 [k16]
+  float* xtbuf = nhc + k16;                      // 2 x 128-row tiles
+  float* xn = xtbuf + 2 * KM_TR2 * d4;           // row norms [KM_TR2]
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;   // 0..7
+  const int lj = lane & 15;
+  const int lk = lane >> 4;
+
+  for (int i = tid; i < k16 * cs; i += KM_NT) ct[i] = 0.0f;
+  __syncthreads();
+  for (int i = tid; i < k * d; i += KM_NT) {
+    const int j = i / d;
+    ct[j * cs + (i - j * d)] = cen[i];
+  }
+  __syncthreads();
+  for (int j = tid; j < k16; j += KM_NT) {
+    if (j < k) {
+      float s = 0.0f;
+      const float* cp = ct + j * cs;
+      for (int kk = 0; kk < d4; ++kk) s += cp[kk] * cp[kk];
+      nhc[j] = -0.5f * s;
+    } else {
+      nhc[j] = -INFINITY;
+    }
+  }
+  __syncthreads();
+
+  const long tiles = (m + KM_TR2 - 1) / KM_TR2;
+  const bool use_glds = (d == d4) && ((PHASES & 1) != 0);
+  int cur = 0;
+  long tile = blockIdx.x;
+
+  auto stage_direct = [&](float* dst, long t) {
+    const long dr0 = t * KM_TR2;
+    const int drows = (int)min((long)KM_TR2, m - dr0);
+    if (d == d4) {
+      const long total4 = ((long)drows * d) >> 2;
+      const float4* src4 = (const float4*)(x + dr0 * (long)d);
+      float4* dst4 = (float4*)dst;
+      for (long i = tid; i < total4; i += KM_NT) dst4[i] = src4[i];
+    } else {
+      for (int i = tid; i < drows * d; i += KM_NT) {
+        const int rr = i / d;
+        dst[rr * d4 + (i - rr * d)] = x[dr0 * (long)d + i];
+      }
+      for (int i = tid; i < drows * (d4 - d); i += KM_NT) {
+        const int rr = i / (d4 - d);
+        dst[rr * d4 + d + (i - rr * (d4 - d))] = 0.0f;
+      }
+    }
+  };
+  auto glds8 = [&](float* dst, long t) {
+    glds_tile_w(x + t * KM_TR2 * (long)d, dst, KM_TR2 * d4, wave, lane, 8);
+  };
+
+  if ((PHASES & 1) && tile < tiles) {
+    if (use_glds && m - tile * KM_TR2 >= KM_TR2) {
+      glds8(xtbuf, tile);
+      vm_drain();
+    } else {
+      stage_direct(xtbuf, tile);
+    }
+  }
+  lds_barrier();
+  for (; tile < tiles; tile += gridDim.x) {
+    const long r0 = tile * KM_TR2;
+    const int rows = (int)min((long)KM_TR2, m - r0);
+    float* xt = xtbuf + cur * KM_TR2 * d4;
+    const long ntile = tile + gridDim.x;
+    const bool pf = use_glds && ntile < tiles &&
+                    (m - ntile * KM_TR2) >= KM_TR2;
+    if (pf) glds8(xtbuf + (cur ^ 1) * KM_TR2 * d4, ntile);
+    if (PHASES & 2) {
+      // row norms: 4 lanes per row over the 128-row tile
+      {
+        const int r = tid >> 2;
+        const int q = tid & 3;
+        float s = 0.0f;
+        if (r < rows) {
+          const float* xp = xt + r * d4;
+          for (int j = q; j < d4; j += 4) s += xp[j] * xp[j];
+        }
+        s += __shfl_xor(s, 1, 4);
+        s += __shfl_xor(s, 2, 4);
+        if (r < rows && q == 0) xn[r] = s;
+      }
+      lds_barrier();
+      const int ri = wave * 16 + lj;
+      const float* xrow = xt + ri * d4 + lk;
+      const float* cb = ct + lj * cs + lk;
+      const int kl = d4 - 4;
+      float best0 = -INFINITY, best1 = -INFINITY, best2 = -INFINITY,
+            best3 = -INFINITY;
+      int bj0 = 0, bj1 = 0, bj2 = 0, bj3 = 0;
+      {
+        f32x4 acc[NTILES];
+#pragma unroll
+        for (int t = 0; t < NTILES; ++t) {
+          const float v = nhc[16 * t + lj];
+          acc[t] = (f32x4){v, v, v, v};
+        }
+        float a0 = xrow[0];
+        float a1 = xrow[min(4, kl)];
+        float b0[NTILES], b1[NTILES];
+#pragma unroll
+        for (int t = 0; t < NTILES; ++t) {
+          b0[t] = cb[16 * t * cs];
+          b1[t] = cb[16 * t * cs + min(4, kl)];
+        }
+        int k0 = 0;
+        for (; k0 + 8 <= d4; k0 += 8) {
+#pragma unroll
+          for (int t = 0; t < NTILES; ++t)
+            acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t],
+                                                          acc[t], 0, 0, 0);
+#pragma unroll
+          for (int t = 0; t < NTILES; ++t)
+            acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1[t],
+                                                          acc[t], 0, 0, 0);
+          const int kn0 = min(k0 + 8, kl);
+          const int kn1 = min(k0 + 12, kl);
+          a0 = xrow[kn0];
+          a1 = xrow[kn1];
+#pragma unroll
+          for (int t = 0; t < NTILES; ++t) {
+            b0[t] = cb[16 * t * cs + kn0];
+            b1[t] = cb[16 * t * cs + kn1];
+          }
+        }
+        if (k0 < d4) {
+#pragma unroll
+          for (int t = 0; t < NTILES; ++t)
+            acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0[t],
+                                                          acc[t], 0, 0, 0);
+        }
+#pragma unroll
+        for (int t = 0; t < NTILES; ++t) {
+          const int j = 16 * t + lj;
+          if (acc[t][0] > best0) { best0 = acc[t][0]; bj0 = j; }
+          if (acc[t][1] > best1) { best1 = acc[t][1]; bj1 = j; }
+          if (acc[t][2] > best2) { best2 = acc[t][2]; bj2 = j; }
+          if (acc[t][3] > best3) { best3 = acc[t][3]; bj3 = j; }
+        }
+      }
+      if (PHASES & 8) {
+#define KM2_RED(S)                                                          \
+  {                                                                         \
+    const float o0 = __shfl_xor(best0, S, 16);                              \
+    const int j0 = __shfl_xor(bj0, S, 16);                                  \
+    if (o0 > best0 || (o0 == best0 && j0 < bj0)) { best0 = o0; bj0 = j0; }  \
+    const float o1 = __shfl_xor(best1, S, 16);                              \
+    const int j1 = __shfl_xor(bj1, S, 16);                                  \
+    if (o1 > best1 || (o1 == best1 && j1 < bj1)) { best1 = o1; bj1 = j1; }  \
+    const float o2 = __shfl_xor(best2, S, 16);                              \
+    const int j2 = __shfl_xor(bj2, S, 16);                                  \
+    if (o2 > best2 || (o2 == best2 && j2 < bj2)) { best2 = o2; bj2 = j2; }  \
+    const float o3 = __shfl_xor(best3, S, 16);                              \
+    const int j3 = __shfl_xor(bj3, S, 16);                                  \
+    if (o3 > best3 || (o3 == best3 && j3 < bj3)) { best3 = o3; bj3 = j3; }  \
+  }
+        KM2_RED(1)
+        KM2_RED(2)
+        KM2_RED(4)
+        KM2_RED(8)
+#undef KM2_RED
+        if (lj == 0) {
+          const int rbase = wave * 16 + lk * 4;
+          const float bests[4] = {best0, best1, best2, best3};
+          const int bjs[4] = {bj0, bj1, bj2, bj3};
+#pragma unroll
+          for (int g = 0; g < 4; ++g) {
+            const int r = rbase + g;
+            if (r < rows) {
+              labels[r0 + r] = bjs[g];
+              dists[r0 + r] = fmaxf(xn[r] - 2.0f * bests[g], 0.0f);
+            }
+          }
+        }
+      }
+    }
+    if (pf) {
+      vm_drain();
+    } else if ((PHASES & 1) && ntile < tiles) {
+      stage_direct(xtbuf + (cur ^ 1) * KM_TR2 * d4, ntile);
+    }
+    lds_barrier();
+    cur ^= 1;
+  }
+}
+
+static size_t closest128_smem_bytes(int k, int d) {
+  const int d4 = (d + 3) & ~3;
+  const int cs = d4 + (((d4 & 7) == 0) ? 4 : 0);
+  const int k16 = (k + 15) & ~15;
+  size_t f32s = (size_t)k16 * cs + k16 + (size_t)2 * KM_TR2 * d4 + KM_TR2;
+  return ((f32s * 4 + 15) & ~15);
+}
+
 static bool kmeans_split_use128(int k, int d) {
   return ((d + 3) & ~3) <= 64 && k <= 128;
 }
@@ -1556,6 +1773,29 @@ void launch_kmeans_probe(const float* x, const float* cen, long m, int d,
 void launch_closest_center(const float* x, const float* cen, long m, int d,
                            int k, int* labels, float* dists, int nblocks,
                            hipStream_t stream) {
+  const char* e = getenv("MI355ML_CC128");
+  const bool allow128 = !(e && atoi(e) == 0);
+  // 128-row tiles win where per-tile fixed costs dominate (small d);
+  // at d~100 they measured a wash (1.448 vs 1.427 ms at 4e6x100)
+  if (allow128 && ((d + 3) & ~3) <= 64 &&
+      closest128_smem_bytes(k, d) <= 160 * 1024 - 1024) {
+    const int ntiles = ((k + 15) & ~15) / 16;
+    const size_t smem = closest128_smem_bytes(k, d);
+#define CC2_LAUNCH(NTL)                                                     \
+  case NTL:                                                                 \
+    hipLaunchKernelGGL((closest_center128_k<NTL>), dim3(nblocks),           \
+                       dim3(KM_NT), smem, stream, x, cen, m, d, k, labels,  \
+                       dists);                                              \
+    break
+    switch (ntiles) {
+      CC2_LAUNCH(1); CC2_LAUNCH(2); CC2_LAUNCH(3); CC2_LAUNCH(4);
+      CC2_LAUNCH(5); CC2_LAUNCH(6); CC2_LAUNCH(7); CC2_LAUNCH(8);
+      CC2_LAUNCH(9); CC2_LAUNCH(10); CC2_LAUNCH(11); CC2_LAUNCH(12);
+      CC2_LAUNCH(13); CC2_LAUNCH(14); CC2_LAUNCH(15); CC2_LAUNCH(16);
+    }
+#undef CC2_LAUNCH
+    return;
+  }
   kmeans_dispatch<0>(x, cen, m, d, k, labels, dists, nullptr, nullptr,
                      nullptr, nblocks, stream);
 }
