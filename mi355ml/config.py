@@ -33,6 +33,12 @@ DEFAULTS: dict = {
         # fall back to eager torch on a GPU box)
         "require_hip": True,
     },
+    "search": {
+        # intra-rank thread parallelism for adaptive-search trials (the
+        # reference runs one sklearn fit per dask worker THREAD; sklearn's
+        # SGD releases the GIL). 0 = serial.
+        "threads": 8,
+    },
     "random": {"seed": 0},
     "logging": {"level": "WARNING"},
 }

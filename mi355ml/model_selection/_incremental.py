@@ -115,22 +115,34 @@ class BaseIncrementalSearchCV(BaseEstimator):
 
         budgets = {t.tid: self._initial_budget() for t in trials}
         round_idx = 0
+
+        def _run_trial(t, n_pf):
+            for _ in range(n_pf):
+                blk = t.pf_calls % len(xs)
+                if classes is not None and t.pf_calls == 0:
+                    t.model.partial_fit(xs[blk], ys[blk],
+                                        classes=classes, **fit_params)
+                else:
+                    t.model.partial_fit(xs[blk], ys[blk], **fit_params)
+                t.pf_calls += 1
+
+        nthreads = int(config.get("search.threads") or 0)
         while budgets:
-            # run this round's partial_fits for my trials
-            stepped = {}
-            for tid, n_pf in budgets.items():
-                t = owned.get(tid)
-                if t is None:
-                    continue
-                for _ in range(n_pf):
-                    blk = t.pf_calls % len(xs)
-                    if classes is not None and t.pf_calls == 0:
-                        t.model.partial_fit(xs[blk], ys[blk],
-                                            classes=classes, **fit_params)
-                    else:
-                        t.model.partial_fit(xs[blk], ys[blk], **fit_params)
-                    t.pf_calls += 1
-                stepped[tid] = (t.pf_calls, t.model)
+            # run this round's partial_fits for my trials — trials are
+            # independent, so they fan out over an intra-rank thread pool
+            # (the reference runs one sklearn fit per dask worker THREAD;
+            # sklearn's SGD releases the GIL). Results are identical to
+            # serial execution: no state is shared between trials.
+            mine = [(tid, owned[tid], n_pf) for tid, n_pf in budgets.items()
+                    if tid in owned]
+            if nthreads > 1 and len(mine) > 1:
+                from concurrent.futures import ThreadPoolExecutor
+                with ThreadPoolExecutor(min(nthreads, len(mine))) as pool:
+                    list(pool.map(lambda a: _run_trial(a[1], a[2]), mine))
+            else:
+                for _, t, n_pf in mine:
+                    _run_trial(t, n_pf)
+            stepped = {tid: (t.pf_calls, t.model) for tid, t, _ in mine}
             # Score on the FULL validation set (reference parity: every
             # model is judged on the same data — VERDICT r1 #7): ship the
             # tiny models to every rank, score each on the local val
@@ -145,10 +157,19 @@ class BaseIncrementalSearchCV(BaseEstimator):
             else:
                 models = stepped
             nval = len(yval)
-            local = {
-                tid: ((float(m.score(xval, yval)) * nval) if nval else 0.0,
-                      nval, pf)
-                for tid, (pf, m) in models.items()}
+
+            def _score_one(item):
+                tid, (pf, m) = item
+                sc = (float(m.score(xval, yval)) * nval) if nval else 0.0
+                return tid, (sc, nval, pf)
+
+            items = list(models.items())
+            if nthreads > 1 and len(items) > 1:
+                from concurrent.futures import ThreadPoolExecutor
+                with ThreadPoolExecutor(min(nthreads, len(items))) as pool:
+                    local = dict(pool.map(_score_one, items))
+            else:
+                local = dict(map(_score_one, items))
             if comm.world_size > 1:
                 agg: Dict[int, tuple] = {}
                 for part in comm.all_gather_object(local):

@@ -388,3 +388,26 @@ def _stealing_world2(rank, world, n):
 
 def test_trial_stealing_rebalances_world2(mp_runner):
     mp_runner(_stealing_world2, world=2, args=(600,))
+
+
+def test_search_threads_identical_to_serial():
+    # trial fan-out over threads must be bit-identical to serial (no
+    # shared state between trials)
+    import mi355ml.config as config
+    X, y = _cls(n=1500, seed=16)
+    def run():
+        s = IncrementalSearchCV(
+            SGDClassifier(loss="log_loss", random_state=0),
+            {"alpha": [1e-5, 1e-4, 1e-3, 1e-2]},
+            n_initial_parameters=4, max_iter=6, random_state=0)
+        s.fit(from_numpy(X), from_numpy(y))
+        return s
+    with config.set({"search.threads": 0}):
+        a = run()
+    with config.set({"search.threads": 8}):
+        b = run()
+    assert a.best_params_ == b.best_params_
+    np.testing.assert_array_equal(a.cv_results_["test_score"],
+                                  b.cv_results_["test_score"])
+    np.testing.assert_array_equal(a.best_estimator_.coef_,
+                                  b.best_estimator_.coef_)
