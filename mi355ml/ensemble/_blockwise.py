@@ -26,17 +26,30 @@ class _BlockwiseBase(BaseEstimator):
 
     def _fit(self, X: DistArray, y: DistArray, **kwargs):
         bs = int(config.get("array.block_rows"))
-        local_models = []
         x = X.local
         yl = y.local
+        blocks = []
         for lo in range(0, max(x.shape[0], 1), bs):
             xb = x[lo:lo + bs]
-            yb = yl[lo:lo + bs]
-            if xb.shape[0] == 0:
-                continue
+            if xb.shape[0]:
+                blocks.append((xb.cpu().numpy(),
+                               yl[lo:lo + bs].cpu().numpy()))
+
+        def _fit_block(args):
+            xb, yb = args
             est = clone(self.estimator)
-            est.fit(xb.cpu().numpy(), yb.cpu().numpy(), **kwargs)
-            local_models.append(est)
+            est.fit(xb, yb, **kwargs)
+            return est
+
+        # block fits are independent sklearn CPU fits — fan out over an
+        # intra-rank thread pool (same knob as the adaptive searches)
+        nthreads = int(config.get("search.threads") or 0)
+        if nthreads > 1 and len(blocks) > 1:
+            from concurrent.futures import ThreadPoolExecutor
+            with ThreadPoolExecutor(min(nthreads, len(blocks))) as pool:
+                local_models = list(pool.map(_fit_block, blocks))
+        else:
+            local_models = [_fit_block(b) for b in blocks]
         gathered = X.comm.all_gather_object(local_models)
         self.estimators_ = [m for part in gathered for m in part]
         self.n_features_in_ = X.ncols
