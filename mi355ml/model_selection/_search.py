@@ -208,9 +208,24 @@ class _SearchBase(BaseEstimator):
             ytr = tr[1] if yl is not None else None
             yte = te[1] if yl is not None else None
             cache: Dict[str, Any] = {}
-            for c in mine:
-                ts, tr_s = self._fit_candidate(
+
+            def _one(c):
+                return c, self._fit_candidate(
                     candidates[c], Xtr, ytr, Xte, yte, cache, f, scorers)
+
+            # candidate fits on the locally-gathered data are independent
+            # (world-1 communicator, no collectives) — fan them out over
+            # the intra-rank thread pool. The prefix cache is shared:
+            # worst-case race is a duplicated prefix fit, never a wrong one
+            from .. import config as _config
+            nthreads = int(_config.get("search.threads") or 0)
+            if nthreads > 1 and len(mine) > 1:
+                from concurrent.futures import ThreadPoolExecutor
+                with ThreadPoolExecutor(min(nthreads, len(mine))) as pool:
+                    results = list(pool.map(_one, mine))
+            else:
+                results = [_one(c) for c in mine]
+            for c, (ts, tr_s) in results:
                 for n in scorers:
                     test[n][c, f] = ts[n]
                     if train is not None:
