@@ -145,3 +145,15 @@ def set(values: dict | None = None, **kwargs: Any) -> Iterator[None]:
 
 
 refresh()
+
+
+def capture_overlays() -> list:
+    """Snapshot this thread's config.set overlays (for propagating the
+    active config context into worker threads — overlays are
+    thread-local)."""
+    return [copy.deepcopy(o) for o in _overlays()]
+
+
+def install_overlays(overlays: list) -> None:
+    """Install a snapshot from capture_overlays() in the current thread."""
+    _local.overlays = list(overlays)

@@ -46,8 +46,14 @@ class _BlockwiseBase(BaseEstimator):
         nthreads = int(config.get("search.threads") or 0)
         if nthreads > 1 and len(blocks) > 1:
             from concurrent.futures import ThreadPoolExecutor
+            ovs = config.capture_overlays()
+
+            def _fit_w(b):
+                config.install_overlays(ovs)
+                return _fit_block(b)
+
             with ThreadPoolExecutor(min(nthreads, len(blocks))) as pool:
-                local_models = list(pool.map(_fit_block, blocks))
+                local_models = list(pool.map(_fit_w, blocks))
         else:
             local_models = [_fit_block(b) for b in blocks]
         gathered = X.comm.all_gather_object(local_models)

@@ -137,8 +137,14 @@ class BaseIncrementalSearchCV(BaseEstimator):
                     if tid in owned]
             if nthreads > 1 and len(mine) > 1:
                 from concurrent.futures import ThreadPoolExecutor
+                ovs = config.capture_overlays()
+
+                def _worker(a):
+                    config.install_overlays(ovs)
+                    return _run_trial(a[1], a[2])
+
                 with ThreadPoolExecutor(min(nthreads, len(mine))) as pool:
-                    list(pool.map(lambda a: _run_trial(a[1], a[2]), mine))
+                    list(pool.map(_worker, mine))
             else:
                 for _, t, n_pf in mine:
                     _run_trial(t, n_pf)
@@ -166,8 +172,14 @@ class BaseIncrementalSearchCV(BaseEstimator):
             items = list(models.items())
             if nthreads > 1 and len(items) > 1:
                 from concurrent.futures import ThreadPoolExecutor
+                ovs = config.capture_overlays()
+
+                def _score_worker(item):
+                    config.install_overlays(ovs)
+                    return _score_one(item)
+
                 with ThreadPoolExecutor(min(nthreads, len(items))) as pool:
-                    local = dict(pool.map(_score_one, items))
+                    local = dict(pool.map(_score_worker, items))
             else:
                 local = dict(map(_score_one, items))
             if comm.world_size > 1:

@@ -221,8 +221,14 @@ class _SearchBase(BaseEstimator):
             nthreads = int(_config.get("search.threads") or 0)
             if nthreads > 1 and len(mine) > 1:
                 from concurrent.futures import ThreadPoolExecutor
+                ovs = _config.capture_overlays()
+
+                def _one_w(c):
+                    _config.install_overlays(ovs)
+                    return _one(c)
+
                 with ThreadPoolExecutor(min(nthreads, len(mine))) as pool:
-                    results = list(pool.map(_one, mine))
+                    results = list(pool.map(_one_w, mine))
             else:
                 results = [_one(c) for c in mine]
             for c, (ts, tr_s) in results:
