@@ -24,9 +24,12 @@ ArrayLike = Union[DistArray, np.ndarray, torch.Tensor]
 
 
 def check_array(X: ArrayLike, dtype: torch.dtype = torch.float32,
-                comm=None) -> DistArray:
+                comm=None, col_ok: bool = False) -> DistArray:
     """Coerce input to a row-sharded DistArray (dask-tolerant check_array
-    analog, dask_ml/utils.py)."""
+    analog, dask_ml/utils.py). Column-sharded (ColDistArray) inputs pass
+    through ONLY when the caller opts in with col_ok=True — estimators
+    that assume row shards would silently misread a column slice as a
+    row shard otherwise."""
     if isinstance(X, DistArray):
         if dtype is not None and X.dtype != dtype:
             from .core.spill import SpilledDistArray
@@ -37,6 +40,12 @@ def check_array(X: ArrayLike, dtype: torch.dtype = torch.float32,
         return X
     from .core.colarray import ColDistArray, Replicated
     if isinstance(X, ColDistArray):
+        if not col_ok:
+            raise TypeError(
+                "column-sharded (ColDistArray) input is only supported by "
+                "the feature-parallel paths (GLM lbfgs/GD/prox, "
+                "TruncatedSVD, PCA); this estimator needs row-sharded "
+                "data — use mi355ml.from_numpy / shard_rows")
         # column-sharded (wide-X) arrays pass through untouched
         if dtype is not None and X.dtype != dtype:
             return X.astype(dtype)
@@ -61,8 +70,9 @@ def check_array(X: ArrayLike, dtype: torch.dtype = torch.float32,
 
 
 def check_X_y(X: ArrayLike, y: ArrayLike, dtype: torch.dtype = torch.float32,
-              y_dtype: torch.dtype = torch.float32, comm=None):
-    Xd = check_array(X, dtype, comm)
+              y_dtype: torch.dtype = torch.float32, comm=None,
+              col_ok: bool = False):
+    Xd = check_array(X, dtype, comm, col_ok=col_ok)
     from .core.colarray import ColDistArray, Replicated
     if isinstance(Xd, ColDistArray):
         # col-sharded X holds ALL rows on every rank: y must be the full

@@ -115,12 +115,12 @@ class PCA(BaseEstimator, TransformerMixin):
         return Uk, s, Vh
 
     def fit(self, X, y=None):
-        X = check_array(X)
+        X = check_array(X, col_ok=True)
         self._fit(X)
         return self
 
     def fit_transform(self, X, y=None):
-        X = check_array(X)
+        X = check_array(X, col_ok=True)
         U, s, _ = self._fit(X, need_u=True)
         if self.whiten:
             import math
@@ -130,7 +130,7 @@ class PCA(BaseEstimator, TransformerMixin):
 
     def transform(self, X):
         check_is_fitted(self, "components_")
-        X = check_array(X)
+        X = check_array(X, col_ok=True)
         V = self.components_.to(X.device)
         mean = self.mean_.to(X.device)
         from ..core.colarray import ColDistArray, Replicated

@@ -70,17 +70,17 @@ class TruncatedSVD(BaseEstimator, TransformerMixin):
         return U, s, Vh
 
     def fit(self, X, y=None):
-        self._fit(check_array(X))
+        self._fit(check_array(X, col_ok=True))
         return self
 
     def fit_transform(self, X, y=None):
-        X = check_array(X)
+        X = check_array(X, col_ok=True)
         U, s, _ = self._fit(X, need_u=True)
         return U.map_local(lambda t: t * s.to(t.dtype))
 
     def transform(self, X):
         check_is_fitted(self, "components_")
-        X = check_array(X)
+        X = check_array(X, col_ok=True)
         V = self.components_.to(X.device)
         from ..core.colarray import ColDistArray, Replicated
         if isinstance(X, ColDistArray):

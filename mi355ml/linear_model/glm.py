@@ -38,7 +38,7 @@ class _GLMBase(BaseEstimator):
         self.eta0 = eta0
 
     def fit(self, X, y):
-        X, y = check_X_y(X, y)
+        X, y = check_X_y(X, y, col_ok=True)
         if self.solver not in SOLVERS:
             raise ValueError(f"unknown solver {self.solver!r}; "
                              f"choose from {sorted(SOLVERS)}")
@@ -73,7 +73,7 @@ class _GLMBase(BaseEstimator):
         usable under wrappers.Incremental and the Hyperband family."""
         import math
         from ..glm.algorithms import _dist_value_grad
-        X, y = check_X_y(X, y)
+        X, y = check_X_y(X, y, col_ok=True)
         d = X.ncols
         dev = X.device
         if not hasattr(self, "coef_") or self.coef_ is None:
@@ -96,7 +96,7 @@ class _GLMBase(BaseEstimator):
 
     def _linear(self, X) -> DistArray:
         check_is_fitted(self, "coef_")
-        X = check_array(X)
+        X = check_array(X, col_ok=True)
         w = self.coef_.to(X.device)
         b = self.intercept_
         from ..core.sparse import SparseDistArray

@@ -180,3 +180,23 @@ def test_col_pca_world2(mp_runner):
     V = rng.standard_normal((4, 60)).astype(np.float32)
     X = (U @ V + 0.05 * rng.standard_normal((400, 60))).astype(np.float32)
     mp_runner(_col_pca_world2, world=2, args=(X.tobytes(), 400, 60))
+
+
+def test_row_sharded_estimators_reject_col_input():
+    """Row-shard-only estimators must refuse ColDistArray loudly — a
+    column slice silently treated as a row shard would be WRONG at
+    world > 1."""
+    from mi355ml.cluster import KMeans
+    from mi355ml.model_selection import GridSearchCV, train_test_split
+    from mi355ml.preprocessing import StandardScaler
+    X, y = _data(n=100, d=8)
+    C = from_numpy_cols(X)
+    with pytest.raises(TypeError, match="column-sharded"):
+        KMeans(n_clusters=2, max_iter=2).fit(C)
+    with pytest.raises(TypeError, match="column-sharded"):
+        StandardScaler().fit(C)
+    with pytest.raises(TypeError, match="column-sharded"):
+        train_test_split(C, torch.from_numpy(y))
+    with pytest.raises(TypeError, match="column-sharded"):
+        GridSearchCV(LogisticRegression(), {"C": [1.0]}, cv=2).fit(
+            C, torch.from_numpy(y))
