@@ -1,5 +1,9 @@
 """Hyperband over SGDClassifier.partial_fit (BASELINE config 5):
 trials fan out over the ranks, data stays sharded in place."""
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.join(_os.path.dirname(
+    _os.path.abspath(__file__)), ".."))
 from sklearn.linear_model import SGDClassifier
 
 from mi355ml import datasets

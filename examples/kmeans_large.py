@@ -3,6 +3,10 @@
 8 GPUs: python -m mi355ml.parallel.launch --gpus 8 examples/kmeans_large.py
 (1e9 x 50 fp32 = 25 GB per GPU at 8 ranks.)
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.join(_os.path.dirname(
+    _os.path.abspath(__file__)), ".."))
 import time
 
 from mi355ml import datasets
@@ -10,7 +14,9 @@ from mi355ml.cluster import KMeans
 from mi355ml.parallel import comm
 
 c = comm.init()
-n = 125_000_000 * c.world_size  # 1e9 at 8 ranks
+# 1e9 at 8 GPU ranks; small smoke size on CPU-only machines
+per_rank = 125_000_000 if c.device.type == "cuda" else 100_000
+n = per_rank * c.world_size
 X, _ = datasets.make_blobs(n_samples=n, n_features=50, centers=100,
                            random_state=0)
 t0 = time.perf_counter()

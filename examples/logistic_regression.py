@@ -3,6 +3,10 @@
 1 GPU:   python examples/logistic_regression.py
 8 GPUs:  python -m mi355ml.parallel.launch --gpus 8 examples/logistic_regression.py
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.join(_os.path.dirname(
+    _os.path.abspath(__file__)), ".."))
 from mi355ml import datasets
 from mi355ml.linear_model import LogisticRegression
 from mi355ml.model_selection import train_test_split

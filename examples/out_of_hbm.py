@@ -1,5 +1,9 @@
 """Beyond-HBM training: the shard lives in pinned host memory and streams
 through the GPU block by block (SpilledDistArray)."""
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.join(_os.path.dirname(
+    _os.path.abspath(__file__)), ".."))
 import torch
 
 from mi355ml.core.array import shard_rows

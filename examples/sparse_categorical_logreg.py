@@ -10,6 +10,10 @@ Run on 1 GPU:    python examples/sparse_categorical_logreg.py
 Run on 8 GPUs:   python -m mi355ml.parallel.launch --gpus 8 \
                      examples/sparse_categorical_logreg.py
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.join(_os.path.dirname(
+    _os.path.abspath(__file__)), ".."))
 
 import numpy as np
 

@@ -1,5 +1,9 @@
 """Text pipeline: HashingVectorizer -> Incremental(SGD) over sparse
 partitions, then serve the model over HTTP."""
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.join(_os.path.dirname(
+    _os.path.abspath(__file__)), ".."))
 import numpy as np
 from sklearn.linear_model import SGDClassifier
 
