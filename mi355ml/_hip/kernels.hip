@@ -158,6 +158,65 @@ __device__ __forceinline__ void glds_tile(const float* __restrict__ src,
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+// Depth-2 read-ahead LDS RMW chain, 8 rows per call (update pass): the
+// C-level bins[l_r*d+c] += v_r chain pays a full LDS round trip per row
+// because the compiler must order read(r+1) after write(r) (may-alias).
+// In hardware the LDS queue is in-order, so read(r+2) may legally issue
+// before write(r+1) whenever l(r+2) != l(r+1) — the caller guarantees
+// ALL ADJACENT labels in the batch are distinct (wave-uniform check,
+// ~93% of batches at k=100; serial fallback otherwise). Two reads stay
+// in flight, hiding most of the per-row latency. Per-column add order is
+// unchanged -> bitwise-identical sums.
+__device__ __forceinline__ void ds_chain_rmw8(
+    unsigned a0, unsigned a1, unsigned a2, unsigned a3,
+    unsigned a4, unsigned a5, unsigned a6, unsigned a7,
+    float v0, float v1, float v2, float v3,
+    float v4, float v5, float v6, float v7) {
+  float o0, o1, o2, o3, o4, o5, o6, o7;
+  asm volatile(
+      "s_waitcnt lgkmcnt(0)\n\t"   // clean queue -> exact counts below
+      "ds_read_b32 %0, %8\n\t"
+      "ds_read_b32 %1, %9\n\t"
+      "s_waitcnt lgkmcnt(1)\n\t"
+      "v_add_f32 %0, %0, %16\n\t"
+      "ds_write_b32 %8, %0\n\t"
+      "ds_read_b32 %2, %10\n\t"
+      "s_waitcnt lgkmcnt(2)\n\t"
+      "v_add_f32 %1, %1, %17\n\t"
+      "ds_write_b32 %9, %1\n\t"
+      "ds_read_b32 %3, %11\n\t"
+      "s_waitcnt lgkmcnt(2)\n\t"
+      "v_add_f32 %2, %2, %18\n\t"
+      "ds_write_b32 %10, %2\n\t"
+      "ds_read_b32 %4, %12\n\t"
+      "s_waitcnt lgkmcnt(2)\n\t"
+      "v_add_f32 %3, %3, %19\n\t"
+      "ds_write_b32 %11, %3\n\t"
+      "ds_read_b32 %5, %13\n\t"
+      "s_waitcnt lgkmcnt(2)\n\t"
+      "v_add_f32 %4, %4, %20\n\t"
+      "ds_write_b32 %12, %4\n\t"
+      "ds_read_b32 %6, %14\n\t"
+      "s_waitcnt lgkmcnt(2)\n\t"
+      "v_add_f32 %5, %5, %21\n\t"
+      "ds_write_b32 %13, %5\n\t"
+      "ds_read_b32 %7, %15\n\t"
+      "s_waitcnt lgkmcnt(2)\n\t"
+      "v_add_f32 %6, %6, %22\n\t"
+      "ds_write_b32 %14, %6\n\t"
+      "s_waitcnt lgkmcnt(1)\n\t"
+      "v_add_f32 %7, %7, %23\n\t"
+      "ds_write_b32 %15, %7"
+      : "=&v"(o0), "=&v"(o1), "=&v"(o2), "=&v"(o3),
+        "=&v"(o4), "=&v"(o5), "=&v"(o6), "=&v"(o7)
+      : "v"(a0), "v"(a1), "v"(a2), "v"(a3),
+        "v"(a4), "v"(a5), "v"(a6), "v"(a7),
+        "v"(v0), "v"(v1), "v"(v2), "v"(v3),
+        "v"(v4), "v"(v5), "v"(v6), "v"(v7)
+      : "memory");
+}
+
+
 
 // ---------------------------------------------------------------------------
 // K1+K2+K3: fused GLM objective + gradient.
@@ -847,21 +906,35 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split_k(
   for (int c = uw * 64 + lane; c < d; c += 128) {                           \
     float sx2 = 0.0f;                                                       \
     const float* xp = xprev + c;                                            \
+    const unsigned bc = (unsigned)(unsigned long long)                      \
+        (__attribute__((address_space(3))) float*)(void*)bins +             \
+        4u * (unsigned)c;                                                   \
     int r = (RA);                                                           \
-    for (; r + 4 <= (RB); r += 4) {                                         \
-      const int l0 = min(max(rl_prev[r], 0), k - 1);                        \
-      const int l1 = min(max(rl_prev[r + 1], 0), k - 1);                    \
-      const int l2 = min(max(rl_prev[r + 2], 0), k - 1);                    \
-      const int l3 = min(max(rl_prev[r + 3], 0), k - 1);                    \
-      const float v0 = xp[(long)r * d];                                     \
-      const float v1 = xp[(long)(r + 1) * d];                               \
-      const float v2 = xp[(long)(r + 2) * d];                               \
-      const float v3 = xp[(long)(r + 3) * d];                               \
-      bins[l0 * d + c] += v0;                                               \
-      bins[l1 * d + c] += v1;                                               \
-      bins[l2 * d + c] += v2;                                               \
-      bins[l3 * d + c] += v3;                                               \
-      sx2 += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;                         \
+    for (; r + 8 <= (RB); r += 8) {                                         \
+      int ls[8];                                                            \
+      float vs[8];                                                          \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                       \
+        ls[j] = min(max(rl_prev[r + j], 0), k - 1);                         \
+        vs[j] = xp[(long)(r + j) * d];                                      \
+        sx2 += vs[j] * vs[j];                                               \
+      }                                                                     \
+      if (ls[0] != ls[1] && ls[1] != ls[2] && ls[2] != ls[3] &&             \
+          ls[3] != ls[4] && ls[4] != ls[5] && ls[5] != ls[6] &&             \
+          ls[6] != ls[7]) {                                                 \
+        ds_chain_rmw8(bc + 4u * (unsigned)(ls[0] * d),                      \
+                      bc + 4u * (unsigned)(ls[1] * d),                      \
+                      bc + 4u * (unsigned)(ls[2] * d),                      \
+                      bc + 4u * (unsigned)(ls[3] * d),                      \
+                      bc + 4u * (unsigned)(ls[4] * d),                      \
+                      bc + 4u * (unsigned)(ls[5] * d),                      \
+                      bc + 4u * (unsigned)(ls[6] * d),                      \
+                      bc + 4u * (unsigned)(ls[7] * d),                      \
+                      vs[0], vs[1], vs[2], vs[3],                           \
+                      vs[4], vs[5], vs[6], vs[7]);                          \
+      } else {                                                              \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j)                       \
+          bins[ls[j] * d + c] += vs[j];                                     \
+      }                                                                     \
     }                                                                       \
     for (; r < (RB); ++r) {                                                 \
       const int lr = min(max(rl_prev[r], 0), k - 1);                        \
@@ -1194,21 +1267,35 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split128_k(
   for (int c = uw * 64 + lane; c < d; c += 128) {                           \
     float sx2 = 0.0f;                                                       \
     const float* xp = xprev + c;                                            \
+    const unsigned bc = (unsigned)(unsigned long long)                      \
+        (__attribute__((address_space(3))) float*)(void*)bins +             \
+        4u * (unsigned)c;                                                   \
     int r = (RA);                                                           \
-    for (; r + 4 <= (RB); r += 4) {                                         \
-      const int l0 = min(max(rl_prev[r], 0), k - 1);                        \
-      const int l1 = min(max(rl_prev[r + 1], 0), k - 1);                    \
-      const int l2 = min(max(rl_prev[r + 2], 0), k - 1);                    \
-      const int l3 = min(max(rl_prev[r + 3], 0), k - 1);                    \
-      const float v0 = xp[(long)r * d];                                     \
-      const float v1 = xp[(long)(r + 1) * d];                               \
-      const float v2 = xp[(long)(r + 2) * d];                               \
-      const float v3 = xp[(long)(r + 3) * d];                               \
-      bins[l0 * d + c] += v0;                                               \
-      bins[l1 * d + c] += v1;                                               \
-      bins[l2 * d + c] += v2;                                               \
-      bins[l3 * d + c] += v3;                                               \
-      sx2 += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;                         \
+    for (; r + 8 <= (RB); r += 8) {                                         \
+      int ls[8];                                                            \
+      float vs[8];                                                          \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                       \
+        ls[j] = min(max(rl_prev[r + j], 0), k - 1);                         \
+        vs[j] = xp[(long)(r + j) * d];                                      \
+        sx2 += vs[j] * vs[j];                                               \
+      }                                                                     \
+      if (ls[0] != ls[1] && ls[1] != ls[2] && ls[2] != ls[3] &&             \
+          ls[3] != ls[4] && ls[4] != ls[5] && ls[5] != ls[6] &&             \
+          ls[6] != ls[7]) {                                                 \
+        ds_chain_rmw8(bc + 4u * (unsigned)(ls[0] * d),                      \
+                      bc + 4u * (unsigned)(ls[1] * d),                      \
+                      bc + 4u * (unsigned)(ls[2] * d),                      \
+                      bc + 4u * (unsigned)(ls[3] * d),                      \
+                      bc + 4u * (unsigned)(ls[4] * d),                      \
+                      bc + 4u * (unsigned)(ls[5] * d),                      \
+                      bc + 4u * (unsigned)(ls[6] * d),                      \
+                      bc + 4u * (unsigned)(ls[7] * d),                      \
+                      vs[0], vs[1], vs[2], vs[3],                           \
+                      vs[4], vs[5], vs[6], vs[7]);                          \
+      } else {                                                              \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j)                       \
+          bins[ls[j] * d + c] += vs[j];                                     \
+      }                                                                     \
     }                                                                       \
     for (; r < (RB); ++r) {                                                 \
       const int lr = min(max(rl_prev[r], 0), k - 1);                        \
@@ -1691,21 +1778,35 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split128s_k(
   for (int c = uw * 64 + lane; c < d; c += 128) {                           \
     float sx2 = 0.0f;                                                       \
     const float* xp = xprev + c;                                            \
+    const unsigned bc = (unsigned)(unsigned long long)                      \
+        (__attribute__((address_space(3))) float*)(void*)bins +             \
+        4u * (unsigned)c;                                                   \
     int r = (RA);                                                           \
-    for (; r + 4 <= (RB); r += 4) {                                         \
-      const int l0 = min(max(rl_prev[r], 0), k - 1);                        \
-      const int l1 = min(max(rl_prev[r + 1], 0), k - 1);                    \
-      const int l2 = min(max(rl_prev[r + 2], 0), k - 1);                    \
-      const int l3 = min(max(rl_prev[r + 3], 0), k - 1);                    \
-      const float v0 = xp[(long)r * d];                                     \
-      const float v1 = xp[(long)(r + 1) * d];                               \
-      const float v2 = xp[(long)(r + 2) * d];                               \
-      const float v3 = xp[(long)(r + 3) * d];                               \
-      bins[l0 * d + c] += v0;                                               \
-      bins[l1 * d + c] += v1;                                               \
-      bins[l2 * d + c] += v2;                                               \
-      bins[l3 * d + c] += v3;                                               \
-      sx2 += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;                         \
+    for (; r + 8 <= (RB); r += 8) {                                         \
+      int ls[8];                                                            \
+      float vs[8];                                                          \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j) {                       \
+        ls[j] = min(max(rl_prev[r + j], 0), k - 1);                         \
+        vs[j] = xp[(long)(r + j) * d];                                      \
+        sx2 += vs[j] * vs[j];                                               \
+      }                                                                     \
+      if (ls[0] != ls[1] && ls[1] != ls[2] && ls[2] != ls[3] &&             \
+          ls[3] != ls[4] && ls[4] != ls[5] && ls[5] != ls[6] &&             \
+          ls[6] != ls[7]) {                                                 \
+        ds_chain_rmw8(bc + 4u * (unsigned)(ls[0] * d),                      \
+                      bc + 4u * (unsigned)(ls[1] * d),                      \
+                      bc + 4u * (unsigned)(ls[2] * d),                      \
+                      bc + 4u * (unsigned)(ls[3] * d),                      \
+                      bc + 4u * (unsigned)(ls[4] * d),                      \
+                      bc + 4u * (unsigned)(ls[5] * d),                      \
+                      bc + 4u * (unsigned)(ls[6] * d),                      \
+                      bc + 4u * (unsigned)(ls[7] * d),                      \
+                      vs[0], vs[1], vs[2], vs[3],                           \
+                      vs[4], vs[5], vs[6], vs[7]);                          \
+      } else {                                                              \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j)                       \
+          bins[ls[j] * d + c] += vs[j];                                     \
+      }                                                                     \
     }                                                                       \
     for (; r < (RB); ++r) {                                                 \
       const int lr = min(max(rl_prev[r], 0), k - 1);                        \

@@ -23,9 +23,11 @@ for m, d, k in [(4_000_000, 100, 100), (4_000_000, 50, 100),
     else:
         ok_s = bool((sa == sb).all())
     ok_c = bool((ca == cb).all())
-    # inertia: same f64 partials, different reduction-tree order across
-    # the 640- vs 512-thread red[] -> equal to ~1e-12 rel, not bitwise
-    ok_i = abs(float(ia) - float(ib)) <= 1e-10 * max(abs(float(ia)), 1.0)
+    # inertia: f64 reduction-tree order differs across thread counts AND
+    # the per-lane f32 sx2 accumulation order differs between the 8-row
+    # chain batches and the inline kernel's 4-term expressions -> agree
+    # to ~1e-6 rel (real accumulation bugs show up at %-level)
+    ok_i = abs(float(ia) - float(ib)) <= 1e-6 * max(abs(float(ia)), 1.0)
     print(f"m={m} d={d} k={k}: labels={ok_l} sums={ok_s} cnt={ok_c} inertia={ok_i}")
     assert ok_l and ok_s and ok_c and ok_i
 
