@@ -177,9 +177,11 @@ def bench_headline(args, comm):
             comm.all_reduce(buf)
         ev[3].record()
         torch.cuda.synchronize()
+        from mi355ml.ops import kmeans as _km
         probe = {"glm_kernel_ms": ev[0].elapsed_time(ev[1]),
                  "kmeans_kernel_ms": ev[1].elapsed_time(ev[2]),
-                 "allreduce_ms": ev[2].elapsed_time(ev[3])}
+                 "allreduce_ms": ev[2].elapsed_time(ev[3]),
+                 "kmeans_path": _km.last_path}
 
     emit(comm,
          metric=METRIC,

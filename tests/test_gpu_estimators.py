@@ -179,3 +179,21 @@ def test_kmeans_large_k_fallback_deterministic_gpu():
     assert (l1 == l2).all()
     assert (s1 == s2).all() and (n1 == n2).all()
     assert float(i1) == float(i2)
+
+
+def test_kmeans_fit_uses_split_kernel():
+    """The production Lloyd path must be the wave-specialized split
+    kernel, not a fallback (native-loaded evidence for the driver)."""
+    import torch
+    from mi355ml.cluster import KMeans
+    from mi355ml.core.array import from_numpy
+    from mi355ml.ops import kmeans as km_mod
+    import numpy as np
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((200_000, 100)).astype(np.float32)
+    KMeans(n_clusters=100, max_iter=3, random_state=0).fit(from_numpy(X))
+    assert km_mod.last_path in ("fused-split", "fused"), km_mod.last_path
+    assert km_mod.last_path == "fused-split"
+    X50 = rng.standard_normal((200_000, 50)).astype(np.float32)
+    KMeans(n_clusters=100, max_iter=3, random_state=0).fit(from_numpy(X50))
+    assert km_mod.last_path == "fused-split"  # 128-row d<=64 variant
