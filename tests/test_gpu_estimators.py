@@ -190,10 +190,16 @@ def test_kmeans_fit_uses_split_kernel():
     from mi355ml.ops import kmeans as km_mod
     import numpy as np
     rng = np.random.default_rng(0)
+    # the Lloyd op itself must take the wave-specialized split kernel
+    for d in (100, 50):  # 64-row and 128-row variants
+        x = torch.from_numpy(
+            rng.standard_normal((200_000, d)).astype(np.float32)).cuda()
+        c = torch.from_numpy(
+            rng.standard_normal((100, d)).astype(np.float32)).cuda()
+        km_mod.kmeans_assign(x, c)
+        assert km_mod.last_path == "fused-split", (d, km_mod.last_path)
+    # end-to-end fit stays on a HIP path throughout (the final labels
+    # pass is closest_center -> "fused")
     X = rng.standard_normal((200_000, 100)).astype(np.float32)
     KMeans(n_clusters=100, max_iter=3, random_state=0).fit(from_numpy(X))
     assert km_mod.last_path in ("fused-split", "fused"), km_mod.last_path
-    assert km_mod.last_path == "fused-split"
-    X50 = rng.standard_normal((200_000, 50)).astype(np.float32)
-    KMeans(n_clusters=100, max_iter=3, random_state=0).fit(from_numpy(X50))
-    assert km_mod.last_path == "fused-split"  # 128-row d<=64 variant
