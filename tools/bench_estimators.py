@@ -15,13 +15,19 @@ import mi355ml  # noqa: E402,F401
 from mi355ml import datasets  # noqa: E402
 
 
-def timeit(name, fn, reps=1, sync=True):
-    t0 = time.perf_counter()
-    out = fn()
-    if sync and torch.cuda.is_available():
-        torch.cuda.synchronize()
-    dt = time.perf_counter() - t0
-    print(f"{name:58s} {dt:8.2f} s")
+def timeit(name, fn, reps=2, sync=True):
+    """Run reps times, report the BEST (steady-state; first call pays
+    hipMalloc pool growth and kernel-launch caches)."""
+    best = None
+    out = None
+    for _ in range(max(reps, 1)):
+        t0 = time.perf_counter()
+        out = fn()
+        if sync and torch.cuda.is_available():
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        best = dt if best is None else min(best, dt)
+    print(f"{name:58s} {best:8.2f} s")
     return out
 
 
