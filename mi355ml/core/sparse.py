@@ -308,3 +308,122 @@ def rebalance_sparse(A: SparseDistArray) -> SparseDistArray:
 
 # override the dense p2p rebalance (which needs .local) with the CSR form
 SparseDistArray.rebalance = rebalance_sparse
+
+
+class SpilledSparseDistArray(SparseDistArray):
+    """CSR shard resident in (pinned) host memory, streamed to the device
+    in row blocks (the sparse member of the spill tier — core/spill.py's
+    dense design applied to the (crow, col, vals) triple). SpMV-consuming
+    code (GLM sparse value+grad, column sums) works unchanged: dot /
+    _xt_dot_local / sum accumulate over the streamed blocks with the same
+    deterministic segment reductions."""
+
+    def __init__(self, crow, col, vals, n_cols, global_rows, row_offset,
+                 comm=None, block_rows=None):
+        super().__init__(crow, col, vals, n_cols, global_rows, row_offset,
+                         comm)
+        assert self.vals.device.type == "cpu", "spilled storage is host"
+        from .. import config as _config
+        self.block_rows = block_rows or int(_config.get("array.block_rows"))
+        self._copy_stream = (torch.cuda.Stream(self.comm.device)
+                             if self.comm.device.type == "cuda" else None)
+
+    @property
+    def device(self):  # compute device, not storage
+        return self.comm.device
+
+    def device_blocks(self):
+        """Yield (row_lo, SparseDistArray device block); double-buffered
+        H2D of the three CSR tensors on a copy stream."""
+        dev = self.comm.device
+        m = self.local_rows
+        bs = self.block_rows
+        starts = list(range(0, max(m, 1), bs)) if m else []
+
+        def host_block(lo):
+            hi = min(m, lo + bs)
+            p0 = int(self.crow[lo])
+            p1 = int(self.crow[hi])
+            return (self.crow[lo:hi + 1] - p0, self.col[p0:p1],
+                    self.vals[p0:p1], hi - lo)
+
+        if dev.type != "cuda":
+            for lo in starts:
+                crow_b, col_b, vals_b, rows = host_block(lo)
+                yield lo, SparseDistArray(crow_b, col_b, vals_b,
+                                          self._ncols, rows, 0, self.comm)
+            return
+        bufs = [None, None]
+        evts = [torch.cuda.Event(), torch.cuda.Event()]
+
+        def _issue(i, slot):
+            crow_b, col_b, vals_b, rows = host_block(starts[i])
+            with torch.cuda.stream(self._copy_stream):
+                bufs[slot] = (crow_b.to(dev, non_blocking=True),
+                              col_b.to(dev, non_blocking=True),
+                              vals_b.to(dev, non_blocking=True), rows)
+                evts[slot].record(self._copy_stream)
+
+        if starts:
+            _issue(0, 0)
+        for i, lo in enumerate(starts):
+            slot = i & 1
+            if i + 1 < len(starts):
+                _issue(i + 1, slot ^ 1)
+            torch.cuda.current_stream(dev).wait_event(evts[slot])
+            crow_b, col_b, vals_b, rows = bufs[slot]
+            yield lo, SparseDistArray(crow_b, col_b, vals_b, self._ncols,
+                                      rows, 0, self.comm)
+            evts[slot].record(torch.cuda.current_stream(dev))
+            self._copy_stream.wait_event(evts[slot])
+
+    # -- streamed SpMV ops -------------------------------------------------
+    def dot(self, w: torch.Tensor) -> "DistArray":
+        dt = torch.promote_types(self.vals.dtype, w.dtype)
+        dev = self.comm.device
+        w = w.to(device=dev, dtype=dt)
+        out = None
+        for lo, blk in self.device_blocks():
+            z = blk.dot(w).local
+            if out is None:
+                shape = (self.local_rows,) + tuple(z.shape[1:])
+                out = torch.zeros(shape, dtype=z.dtype, device=dev)
+            out[lo:lo + z.shape[0]] = z
+        if out is None:
+            out = torch.zeros(0, dtype=dt, device=dev)
+        from .array import DistArray
+        return DistArray(out, self.global_rows, self.row_offset, self.comm)
+
+    def _xt_dot_local(self, v, dtype: torch.dtype = torch.float64):
+        r = v.local if hasattr(v, "local") else v
+        r = r.to(device=self.comm.device, dtype=dtype)
+        g = torch.zeros(self._ncols, dtype=dtype, device=self.comm.device)
+        for lo, blk in self.device_blocks():
+            g += blk._xt_dot_local(r[lo:lo + blk.local_rows], dtype)
+        return g
+
+    def sum(self, axis=0):
+        if axis not in (0, None):
+            raise NotImplementedError("sum supports axis=0")
+        g = torch.zeros(self._ncols, dtype=torch.float64,
+                        device=self.comm.device)
+        for _, blk in self.device_blocks():
+            ones = torch.ones(blk.local_rows, dtype=torch.float64,
+                              device=self.comm.device)
+            g += blk._xt_dot_local(ones)  # local partial; ONE reduce below
+        if self.comm.world_size > 1:
+            self.comm.all_reduce(g)
+        return g
+
+
+def spill_sparse(S: SparseDistArray,
+                 block_rows=None) -> SpilledSparseDistArray:
+    """Move a device CSR shard into (pinned) host memory for streaming."""
+    def host(t):
+        t = t.cpu()
+        if torch.cuda.is_available():
+            t = t.pin_memory()
+        return t
+    return SpilledSparseDistArray(host(S.crow), host(S.col), host(S.vals),
+                                  S._ncols, S.global_rows, S.row_offset,
+                                  S.comm, block_rows)

@@ -115,3 +115,60 @@ def test_spilled_world2(mp_runner):
     y = (X[:, 0] + X[:, 1] > 0).astype(np.float32)
     mp_runner(_spilled_world2, world=2,
               args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1]))
+
+
+def test_spilled_sparse_matches_resident():
+    """Sparse spill tier (TODO r2): streamed-CSR dot/xt_dot/sum and a
+    sparse GLM fit match the device-resident CSR results."""
+    import scipy.sparse as ssp
+    import torch
+    from mi355ml.core.sparse import (SparseDistArray, from_scipy,
+                                     spill_sparse)
+    from mi355ml.linear_model import LogisticRegression
+    rng = np.random.default_rng(9)
+    n, dcols = 4000, 31
+    dense = (rng.uniform(size=(n, dcols)) < 0.2) * \
+        rng.standard_normal((n, dcols))
+    S = from_scipy(ssp.csr_matrix(dense.astype(np.float32)))
+    P = spill_sparse(S, block_rows=700)  # multiple partial blocks
+    w = torch.from_numpy(rng.standard_normal(dcols).astype(np.float32))
+    np.testing.assert_allclose(P.dot(w).gather().cpu().numpy(),
+                               S.dot(w).gather().cpu().numpy(),
+                               rtol=1e-5, atol=1e-5)
+    r = torch.from_numpy(rng.standard_normal(n).astype(np.float32))
+    np.testing.assert_allclose(P.xt_dot(r).cpu().numpy(),
+                               S.xt_dot(r).cpu().numpy(), rtol=1e-10)
+    np.testing.assert_allclose(P.sum().cpu().numpy(),
+                               S.sum().cpu().numpy(), rtol=1e-10)
+    y = (dense @ rng.standard_normal(dcols) > 0).astype(np.float32)
+    m_res = LogisticRegression(solver="lbfgs", max_iter=30).fit(
+        S, torch.from_numpy(y))
+    m_spl = LogisticRegression(solver="lbfgs", max_iter=30).fit(
+        P, torch.from_numpy(y))
+    np.testing.assert_allclose(m_spl.coef_.numpy(), m_res.coef_.numpy(),
+                               rtol=1e-4, atol=1e-4)
+
+
+def _spilled_sparse_world2(rank, world, a_bytes, shape):
+    import scipy.sparse as ssp
+    import torch
+    from mi355ml.core.array import from_numpy
+    from mi355ml.core.sparse import from_scipy, spill_sparse
+    from mi355ml.linear_model import LogisticRegression
+    a = np.frombuffer(a_bytes, dtype=np.float32).reshape(shape).copy()
+    S = from_scipy(ssp.csr_matrix(a))
+    P = spill_sparse(S, block_rows=150)
+    y = (a.sum(axis=1) > 0).astype(np.float32)
+    m_spl = LogisticRegression(solver="lbfgs", max_iter=25).fit(
+        P, from_numpy(y))
+    m_res = LogisticRegression(solver="lbfgs", max_iter=25).fit(
+        S, from_numpy(y))
+    np.testing.assert_allclose(m_spl.coef_.numpy(), m_res.coef_.numpy(),
+                               rtol=1e-4, atol=1e-4)
+
+
+def test_spilled_sparse_world2(mp_runner):
+    rng = np.random.default_rng(10)
+    a = ((rng.uniform(size=(900, 13)) < 0.25) *
+         rng.standard_normal((900, 13))).astype(np.float32)
+    mp_runner(_spilled_sparse_world2, world=2, args=(a.tobytes(), a.shape))
