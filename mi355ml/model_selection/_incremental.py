@@ -2,13 +2,17 @@
 (dask_ml/model_selection/_incremental.py parity — SURVEY §3.5).
 
 The reference keeps one sklearn model per candidate alive on the workers as
-futures and drives rounds of {partial_fit on one scattered block, score on a
-validation block, _adapt}. MI355X mapping: trials are pinned
-round-robin to ranks; each rank partial_fits its trials on slabs of its OWN
-shard (data is already where the trials are — no scatter), scores on its
-local validation shard, and rounds synchronize with one tiny
-all_gather_object of (trial_id, score) pairs; the _adapt decision runs
-redundantly+deterministically on every rank.
+futures and drives rounds of {partial_fit on one scattered block, score on
+the shared validation set, _adapt}. MI355X mapping: trials are
+assigned round-robin to ranks and partial_fit over an intra-rank thread
+pool on slabs of the owner's OWN shard (data is already where the trials
+are — no scatter). Scoring is GLOBAL: the round's (tiny) stepped models
+travel in one all_gather_object, every rank scores every model on its
+local validation shard, and the count-weighted combination gives the exact
+full-validation-set score (reference parity). _adapt runs
+redundantly+deterministically on every rank, after which survivors are
+REBALANCED across ranks by next-round budget (work stealing with zero
+transfer cost — the models already traveled for scoring).
 """
 
 from __future__ import annotations

@@ -6,6 +6,9 @@ centroid sums and a tree-reduce. Here it is one fused HIP
 kernel per shard: the x·Cᵀ tile GEMM runs on f32 MFMA, the argmin rides in
 the epilogue, and per-workgroup centroid sums/counts land in a slab that is
 tree-reduced on-device; the cross-rank finish is one ncclAllReduce (C3).
+The production kernel is wave-specialized (8 MFMA waves + 2 update waves
+software-pipelined one tile apart — docs/KERNELS.md); d<=64 takes a
+128-row-tile in-wave-argmax variant; k>256 runs the chunked-fused path.
 
 dist²(x, c) = ‖x‖² − 2·x·c + ‖c‖², with the ‖x‖² term dropped for argmin
 and added back only for inertia.
