@@ -208,6 +208,12 @@ class BaseIncrementalSearchCV(BaseEstimator):
                     else:
                         t.since_best += 1
             round_idx += 1
+            if self.verbose and comm.rank == 0:
+                import logging
+                best_now = max(all_scores.values(), default=(0.0, 0))[0]
+                logging.getLogger("mi355ml.search").info(
+                    "round %d: %d trials scored, best=%.4f",
+                    round_idx, len(all_scores), best_now)
             budgets = self._adapt(
                 [t for t in trials if t.tid in all_scores], round_idx)
             # Keep every trained model alive (they are small sklearn

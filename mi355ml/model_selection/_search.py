@@ -314,8 +314,21 @@ class _SearchBase(BaseEstimator):
     def predict_proba(self, X):
         return self.best_estimator_.predict_proba(X)
 
+    def predict_log_proba(self, X):
+        return self.best_estimator_.predict_log_proba(X)
+
+    def decision_function(self, X):
+        return self.best_estimator_.decision_function(X)
+
     def transform(self, X):
         return self.best_estimator_.transform(X)
+
+    def inverse_transform(self, X):
+        return self.best_estimator_.inverse_transform(X)
+
+    @property
+    def classes_(self):
+        return self.best_estimator_.classes_
 
     def score(self, X, y=None):
         return self._score(self.best_estimator_, check_array(X, dtype=None),
