@@ -43,7 +43,8 @@ METRIC = ("samples/sec (whole node) KMeans k=100 + LogReg fit, "
 
 
 def build_data(rows_local: int, d: int, k: int, comm, dtype=torch.float32,
-               rows_global: int | None = None):
+               rows_global: int | None = None,
+               row_offset: int | None = None):
     """Synthetic blobs + logistic labels, generated per-shard on device
     (random-init equivalent; no dataset downloads — BASELINE.md rules)."""
     dev = comm.device
@@ -57,8 +58,9 @@ def build_data(rows_local: int, d: int, k: int, comm, dtype=torch.float32,
     y = (torch.sigmoid(x @ w_true)
          > torch.rand(rows_local, generator=sg, device=dev)).to(dtype)
     total = rows_global or rows_local * comm.world_size
-    X = DistArray(x.contiguous(), total, comm.rank * rows_local, comm)
-    Y = DistArray(y.contiguous(), total, comm.rank * rows_local, comm)
+    off = row_offset if row_offset is not None else comm.rank * rows_local
+    X = DistArray(x.contiguous(), total, off, comm)
+    Y = DistArray(y.contiguous(), total, off, comm)
     return X, Y, centers.clone()
 
 
@@ -140,13 +142,14 @@ def emit(comm, *, metric, value, unit, args, ms_per_step, config,
 def bench_headline(args, comm):
     on_gpu = comm.device.type == "cuda"
     rows_global = args.rows or (100_000_000 if on_gpu else 100_000)
-    rows_local = rows_global // comm.world_size
-    if comm.rank < rows_global % comm.world_size:
-        rows_local += 1
+    base, rem = divmod(rows_global, comm.world_size)
+    rows_local = base + (1 if comm.rank < rem else 0)
+    row_offset = comm.rank * base + min(comm.rank, rem)
     d, k = args.features, args.clusters
 
     X, Y, centers = build_data(rows_local, d, k, comm,
-                               rows_global=rows_global)
+                               rows_global=rows_global,
+                               row_offset=row_offset)
     w = torch.zeros(d, dtype=torch.float32, device=comm.device)
     b = 0.0
     lr = 1.0
