@@ -32,6 +32,20 @@ from ..core import random as mlrand
 from ._split import train_test_split
 
 
+def _balance_trials(budgets: Dict[int, int], world: int) -> Dict[int, int]:
+    """Deterministic greedy rebalance: biggest next-round budgets first,
+    each onto the least-loaded rank (ties to the lowest rank). Every rank
+    computes the same assignment from the same inputs — no coordination
+    message needed (the models already traveled for global scoring)."""
+    loads = [0.0] * world
+    assign: Dict[int, int] = {}
+    for tid, n_pf in sorted(budgets.items(), key=lambda kv: (-kv[1], kv[0])):
+        r = min(range(world), key=lambda i: (loads[i], i))
+        loads[r] += n_pf
+        assign[tid] = r
+    return assign
+
+
 class _Trial:
     __slots__ = ("tid", "params", "model", "history", "pf_calls",
                  "best_score", "since_best", "score")
@@ -229,12 +243,8 @@ class BaseIncrementalSearchCV(BaseEstimator):
             # scoring — so rebalancing is always profitable. The
             # assignment is computed redundantly+deterministically.
             if budgets and comm.world_size > 1:
-                loads = [0.0] * comm.world_size
-                for tid, n_pf in sorted(budgets.items(),
-                                        key=lambda kv: (-kv[1], kv[0])):
-                    r = min(range(comm.world_size),
-                            key=lambda i: (loads[i], i))
-                    loads[r] += n_pf
+                assign = _balance_trials(budgets, comm.world_size)
+                for tid, r in assign.items():
                     self._owner_of[tid] = r
                     t = next(tt for tt in trials if tt.tid == tid)
                     if r == comm.rank:

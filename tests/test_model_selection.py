@@ -411,3 +411,16 @@ def test_search_threads_identical_to_serial():
                                   b.cv_results_["test_score"])
     np.testing.assert_array_equal(a.best_estimator_.coef_,
                                   b.best_estimator_.coef_)
+
+
+def test_balance_trials_properties():
+    from mi355ml.model_selection._incremental import _balance_trials
+    budgets = {0: 9, 1: 3, 2: 3, 3: 1, 4: 1, 5: 1}
+    a = _balance_trials(budgets, 2)
+    loads = [sum(budgets[t] for t, r in a.items() if r == i)
+             for i in range(2)]
+    assert abs(loads[0] - loads[1]) <= max(budgets.values())
+    assert a == _balance_trials(budgets, 2)  # deterministic
+    # more ranks than trials: everyone gets at most one
+    b = _balance_trials({0: 2, 1: 2}, 8)
+    assert sorted(b.values()) == [0, 1]
