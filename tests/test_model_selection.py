@@ -424,3 +424,10 @@ def test_balance_trials_properties():
     # more ranks than trials: everyone gets at most one
     b = _balance_trials({0: 2, 1: 2}, 8)
     assert sorted(b.values()) == [0, 1]
+
+
+def test_hyperband_world4(mp_runner):
+    # stealing + global scoring + threaded trials under 4 ranks
+    X, y = _cls(n=1600, seed=18)
+    mp_runner(_hyperband_world2, world=4,
+              args=(X.tobytes(), y.tobytes(), X.shape[0], X.shape[1]))
