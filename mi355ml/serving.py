@@ -196,14 +196,16 @@ def create_app(model: Any, max_batch: int = 65536,
 
 def main(argv: Optional[list] = None):
     import argparse
-
-    import joblib
-    import uvicorn
+    import os
 
     ap = argparse.ArgumentParser(prog="mi355ml-serve")
     ap.add_argument("model", help="joblib file with a fitted estimator")
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--device", type=int, default=None,
+                    help="pin this server process to one GPU (run one "
+                         "replica per GPU behind a load balancer; ports "
+                         "port+device by convention)")
     ap.add_argument("--max-batch", type=int, default=65536,
                     help="largest accepted request batch (rows)")
     ap.add_argument("--token", default=None,
@@ -214,11 +216,18 @@ def main(argv: Optional[list] = None):
     ap.add_argument("--max-wait-ms", type=float, default=2.0,
                     help="micro-batch straggler window (with --batching)")
     args = ap.parse_args(argv)
+    if args.device is not None:
+        # must happen before torch initializes HIP in joblib.load
+        os.environ["HIP_VISIBLE_DEVICES"] = str(args.device)
+        os.environ["CUDA_VISIBLE_DEVICES"] = str(args.device)
+    import joblib
+    import uvicorn
     model = joblib.load(args.model)
+    port = args.port + (args.device or 0)
     uvicorn.run(create_app(model, max_batch=args.max_batch,
                            token=args.token, batching=args.batching,
                            max_wait_ms=args.max_wait_ms),
-                host=args.host, port=args.port)
+                host=args.host, port=port)
 
 
 if __name__ == "__main__":
