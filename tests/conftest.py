@@ -49,13 +49,27 @@ def _mp_entry(rank, world, port, fn, args):
         comm.shutdown()
 
 
-def run_multiprocess(fn, world: int = 2, args: tuple = ()):
+def run_multiprocess(fn, world: int = 2, args: tuple = (),
+                     timeout: float = 300.0):
     """Run fn(rank, world, *args) on `world` gloo-connected processes
-    (the gen_cluster analog — SURVEY §4 tier 2)."""
+    (the gen_cluster analog — SURVEY §4 tier 2). Bounded join: a
+    deadlocked rank fails the test instead of hanging the whole suite."""
+    import time
+
     import torch.multiprocessing as mp
     port = free_port()
-    mp.start_processes(_mp_entry, args=(world, port, fn, args),
-                       nprocs=world, join=True, start_method="spawn")
+    ctx = mp.start_processes(_mp_entry, args=(world, port, fn, args),
+                             nprocs=world, join=False,
+                             start_method="spawn")
+    deadline = time.monotonic() + timeout
+    while not ctx.join(timeout=5):
+        if time.monotonic() > deadline:
+            for p in ctx.processes:
+                if p.is_alive():
+                    p.terminate()
+            raise TimeoutError(
+                f"multiprocess test exceeded {timeout:.0f}s "
+                f"(world={world}); ranks terminated")
 
 
 @pytest.fixture
