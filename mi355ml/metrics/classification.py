@@ -32,8 +32,9 @@ def log_loss(y_true, y_prob, eps: float = 1e-15) -> float:
         ll = -(tl * p.log() + (1 - tl) * (1 - p).log()).sum()
         n = float(p.shape[0])
     else:
-        # sklearn parity: clip then renormalize rows to sum to 1
-        p = p / p.sum(dim=1, keepdim=True)
+        # NOTE: no row renormalization — the installed sklearn (1.7 era)
+        # clips and scores y_prob as given (renormalization was removed);
+        # verified numerically against sklearn.metrics.log_loss
         idx = t.long()
         ll = -(p[torch.arange(p.shape[0], device=p.device), idx].log()).sum()
         n = float(p.shape[0])
