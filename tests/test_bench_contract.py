@@ -70,3 +70,46 @@ def test_bench_config_modes_world2(cfg, world):
     assert d["value"] > 0 and d["ms_per_step"] > 0
     assert d["higher_is_better"] is True
     assert "config" in d and "counters" in d["config"]
+
+
+def test_one_step_math_matches_reference():
+    """The bench's fused step must compute a REAL Lloyd update + a real
+    gradient step (guards against accidentally benchmarking skipped
+    work)."""
+    import numpy as np
+    import torch
+
+    import bench
+    from mi355ml.parallel import comm as comm_mod
+
+    comm = comm_mod.default()
+    rng = np.random.default_rng(3)
+    x = torch.from_numpy(rng.standard_normal((500, 8)).astype(np.float32))
+    y = torch.from_numpy((rng.uniform(size=500) < 0.5).astype(np.float32))
+    c0 = torch.from_numpy(rng.standard_normal((4, 8)).astype(np.float32))
+    from mi355ml.core.array import DistArray
+    X = DistArray(x, 500, 0, comm)
+    Y = DistArray(y, 500, 0, comm)
+    w0 = torch.zeros(8)
+    buf = torch.empty(2 + 8 + 4 * 8 + 4 + 1, dtype=torch.float64)
+    w1, b1, c1, loss, inertia = bench.one_step(X, Y, w0.clone(), 0.0,
+                                               c0.clone(), 1.0, comm, buf)
+    # reference Lloyd update
+    d2 = ((x[:, None, :] - c0[None]) ** 2).sum(-1)
+    lab = d2.argmin(1)
+    c_ref = torch.stack([
+        x[lab == j].mean(0) if (lab == j).any() else torch.zeros(8)
+        for j in range(4)])
+    torch.testing.assert_close(c1, c_ref, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(
+        torch.tensor(inertia, dtype=torch.float64),
+        d2.min(1).values.double().sum(), rtol=1e-5, atol=1e-2)
+    # reference logistic gradient step (lr/n scaling)
+    z = x @ w0
+    r = torch.sigmoid(z) - y
+    gw = x.t() @ r
+    torch.testing.assert_close(w1.double(), (w0 - gw / 500).double(),
+                               rtol=1e-5, atol=1e-6)
+    ref_loss = (torch.nn.functional.softplus(z) - y * z).double().sum()
+    torch.testing.assert_close(torch.tensor(loss, dtype=torch.float64),
+                               ref_loss, rtol=1e-5, atol=1e-3)
