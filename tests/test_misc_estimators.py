@@ -194,3 +194,27 @@ def test_r2_constant_y_true_edge_cases():
     y = np.full(64, 3.0, dtype=np.float32)
     assert r2_score(from_numpy(y), from_numpy(y.copy())) == 1.0
     assert r2_score(from_numpy(y), from_numpy(y + 1.0)) == 0.0
+
+
+def test_utils_functions():
+    from mi355ml.utils import (check_chunks, draw_seed,
+                               handle_zeros_in_scale, svd_flip)
+    import torch
+    # svd_flip: deterministic signs, consistent u/v flip
+    u = torch.tensor([[1.0, -2.0], [3.0, 4.0]])
+    v = torch.tensor([[-1.0, 2.0], [3.0, -4.0]])
+    uf, vf = svd_flip(u.clone(), v.clone())
+    uf2, vf2 = svd_flip(u.clone(), v.clone())
+    assert torch.equal(uf, uf2) and torch.equal(vf, vf2)
+    torch.testing.assert_close(uf @ vf, u @ v)  # product preserved
+    # zeros in scale replaced by 1 (sklearn semantics)
+    s = handle_zeros_in_scale(torch.tensor([0.0, 2.0, 0.0]))
+    assert s.tolist() == [1.0, 2.0, 1.0]
+    # chunks tuple sane
+    r, c = check_chunks(100, 10)
+    assert r >= 1 and c >= 1
+    # draw_seed in range + deterministic per rng state
+    rng = np.random.default_rng(0)
+    a = draw_seed(np.random.default_rng(0))
+    b = draw_seed(np.random.default_rng(0))
+    assert a == b and 0 <= a < 2 ** 31
