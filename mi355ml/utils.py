@@ -38,11 +38,13 @@ def check_chunks(n_samples: int, n_features: int, chunks=None) -> tuple:
 
 
 def draw_seed(rng, low: int = 0, high: int = 2 ** 31 - 1) -> int:
-    """Draw a child seed from a random state / int."""
+    """Draw a child seed from an int / RandomState / Generator."""
     import numpy as np
     rs = np.random.RandomState(rng) if isinstance(rng, (int, type(None))) \
         else rng
-    return int(rs.randint(low, high))
+    if hasattr(rs, "randint"):        # RandomState
+        return int(rs.randint(low, high))
+    return int(rs.integers(low, high))  # np.random.Generator
 
 
 def slice_columns(X, columns):
