@@ -807,9 +807,11 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split_k(
   float* nhc = ct + k16 * cs;                    // -0.5*c^2 [k16]
   float* xtbuf = nhc + k16;                      // 2 x-tile buffers
   int* rlab2 = (int*)(xtbuf + 2 * KM_TR * d4);   // [2][KM_TR] label bufs
-  float* cbest = (float*)(rlab2 + 2 * KM_TR);    // [16][KM_TR+4]
-  int* cbj = (int*)(cbest + 16 * (KM_TR + 4));   // [16][KM_TR+4]
-  float* bins = (float*)(cbj + 16 * (KM_TR + 4));  // [k*d]
+  // candidate (best, idx) pairs packed into u64 words: one b64 LDS op
+  // per publish/finalize access instead of two b32 (same total bytes)
+  unsigned long long* cpk =
+      (unsigned long long*)(rlab2 + 2 * KM_TR);  // [16][KM_TR+4]
+  float* bins = (float*)(cpk + 16 * (KM_TR + 4));  // [k*d]
   float* bcnt = bins + k * d;                    // [k]
   double* red = (double*)(smem_raw + ((((size_t)(k16 * cs + k16 +
       2 * KM_TR * d4 + 2 * KM_TR + 32 * (KM_TR + 4) + k * d + k) * 4 +
@@ -1025,8 +1027,9 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split_k(
           const int bjs[4] = {bj0, bj1, bj2, bj3};
 #pragma unroll
           for (int g = 0; g < 4; ++g) {
-            cbest[slot * (KM_TR + 4) + rbase + g] = bests[g];
-            cbj[slot * (KM_TR + 4) + rbase + g] = bjs[g];
+            cpk[slot * (KM_TR + 4) + rbase + g] =
+                ((unsigned long long)__float_as_uint(bests[g]) << 32) |
+                (unsigned)bjs[g];
           }
         }
       }
@@ -1049,11 +1052,13 @@ __global__ __launch_bounds__(KMS_NT) void kmeans_assign_split_k(
         float bv = -INFINITY;
         int bj = 0;
         if (r < rows) {
-          bv = cbest[g8 * (KM_TR + 4) + r];
-          bj = cbj[g8 * (KM_TR + 4) + r];
+          const unsigned long long p0 = cpk[g8 * (KM_TR + 4) + r];
+          bv = __uint_as_float((unsigned)(p0 >> 32));
+          bj = (int)(unsigned)p0;
           if (nslots == 16) {
-            const float ov = cbest[(g8 + 8) * (KM_TR + 4) + r];
-            const int oj = cbj[(g8 + 8) * (KM_TR + 4) + r];
+            const unsigned long long p1 = cpk[(g8 + 8) * (KM_TR + 4) + r];
+            const float ov = __uint_as_float((unsigned)(p1 >> 32));
+            const int oj = (int)(unsigned)p1;
             if (ov > bv || (ov == bv && oj < bj)) { bv = ov; bj = oj; }
           }
         }
