@@ -218,3 +218,25 @@ def test_utils_functions():
     a = draw_seed(np.random.default_rng(0))
     b = draw_seed(np.random.default_rng(0))
     assert a == b and 0 <= a < 2 ** 31
+
+
+def test_diagnostics_profiler_counters():
+    from mi355ml.diagnostics import Counters, Profiler, device_memory
+    prof = Profiler(use_events=False)
+    with prof.track("opA", rows=10):
+        pass
+    with prof.track("opA"):
+        pass
+    with prof.track("opB"):
+        pass
+    s = prof.summary()
+    assert s["opA"]["n"] == 2 and s["opB"]["n"] == 1
+    assert "opA" in prof.report()
+    c = Counters()
+    c.inc("x.calls")
+    c.inc("x.bytes", 128.0)
+    c.inc("x.calls")
+    snap = c.snapshot()
+    assert snap["x.calls"] == 2 and snap["x.bytes"] == 128.0
+    dm = device_memory()
+    assert isinstance(dm, dict)
