@@ -59,3 +59,9 @@ def test_end_to_end_workflow(tmp_path):
         np.testing.assert_array_equal(np.asarray(r.json()["result"]),
                                       p0[:8])
         assert client.get("/metrics").status_code == 200
+
+
+@pytest.mark.gpu
+def test_end_to_end_workflow_gpu(tmp_path):
+    # same flow with device-resident shards (from_numpy lands on cuda:0)
+    test_end_to_end_workflow(tmp_path)
