@@ -200,3 +200,13 @@ def test_row_sharded_estimators_reject_col_input():
     with pytest.raises(TypeError, match="column-sharded"):
         GridSearchCV(LogisticRegression(), {"C": [1.0]}, cv=2).fit(
             C, torch.from_numpy(y))
+
+
+def test_col_partial_fit_converges():
+    X, y = _data(n=600, d=32, seed=9)
+    m = LogisticRegression(C=100.0, eta0=2.0)
+    C = from_numpy_cols(X)
+    yt = torch.from_numpy(y)
+    for _ in range(60):
+        m.partial_fit(C, yt)
+    assert m.score(C, y) > 0.8
